@@ -1,0 +1,194 @@
+// cpd_amd CPU ops: customized-precision numerics on CPU tensors.
+//
+// The reference emulator is CUDA-only (quant_function.py:24-30 raises on CPU);
+// full CPU support here is both a product feature and the host-side oracle the
+// GPU tests compare against (same quant_core.h compiled for both targets).
+#include <torch/extension.h>
+#include <ATen/Parallel.h>
+
+#include "quant_core.h"
+
+namespace {
+
+using at::Tensor;
+
+void check_f32_contig(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.scalar_type() == at::kFloat, name, " must be float32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(!t.is_cuda(), name, " must be a CPU tensor");
+}
+
+constexpr int64_t kGrain = 1 << 14;
+
+Tensor quantize(const Tensor& x, int64_t man_bits, int64_t exp_bits) {
+  check_f32_contig(x, "x");
+  Tensor out = at::empty_like(x);
+  const float* src = x.const_data_ptr<float>();
+  float* dst = out.mutable_data_ptr<float>();
+  at::parallel_for(0, x.numel(), kGrain, [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e; ++i)
+      dst[i] = cpd::cast_fp(src[i], (int)man_bits, (int)exp_bits);
+  });
+  return out;
+}
+
+Tensor quantize_(Tensor x, int64_t man_bits, int64_t exp_bits) {
+  check_f32_contig(x, "x");
+  float* p = x.mutable_data_ptr<float>();
+  at::parallel_for(0, x.numel(), kGrain, [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e; ++i)
+      p[i] = cpd::cast_fp(p[i], (int)man_bits, (int)exp_bits);
+  });
+  return x;
+}
+
+// acc = Q(acc + inc) elementwise — one hop of the sequential/ring
+// low-precision reduction (dist_util.py:65-67 semantics).
+Tensor qadd_(Tensor acc, const Tensor& inc, int64_t man_bits, int64_t exp_bits) {
+  check_f32_contig(acc, "acc");
+  check_f32_contig(inc, "inc");
+  TORCH_CHECK(acc.numel() == inc.numel(), "size mismatch");
+  float* a = acc.mutable_data_ptr<float>();
+  const float* g = inc.const_data_ptr<float>();
+  at::parallel_for(0, acc.numel(), kGrain, [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e; ++i)
+      a[i] = cpd::cast_fp(a[i] + g[i], (int)man_bits, (int)exp_bits);
+  });
+  return acc;
+}
+
+// Kahan hop: every intermediate rounded (dist_util.py:82-88 semantics).
+Tensor kahan_qadd_(Tensor acc, Tensor comp, const Tensor& inc, int64_t man_bits,
+                   int64_t exp_bits) {
+  check_f32_contig(acc, "acc");
+  check_f32_contig(comp, "comp");
+  check_f32_contig(inc, "inc");
+  TORCH_CHECK(acc.numel() == inc.numel() && comp.numel() == acc.numel(),
+              "size mismatch");
+  float* a = acc.mutable_data_ptr<float>();
+  float* c = comp.mutable_data_ptr<float>();
+  const float* g = inc.const_data_ptr<float>();
+  at::parallel_for(0, acc.numel(), kGrain, [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e; ++i)
+      cpd::kahan_qstep(a[i], c[i], g[i], (int)man_bits, (int)exp_bits);
+  });
+  return acc;
+}
+
+// Per-segment APS max-exponent scan over a flat gradient buffer:
+//   out[s] = ceil(log2(max_i |x_i| * world_size)) for segment s,
+//   -100 when the segment is all-zero (mix.py:260 sentinel).
+Tensor seg_max_exp(const Tensor& flat, const Tensor& offsets,
+                   int64_t world_size) {
+  check_f32_contig(flat, "flat");
+  TORCH_CHECK(offsets.scalar_type() == at::kLong && offsets.is_contiguous());
+  const int64_t S = offsets.numel() - 1;
+  Tensor out = at::empty({S}, flat.options());
+  const float* x = flat.const_data_ptr<float>();
+  const int64_t* ofs = offsets.const_data_ptr<int64_t>();
+  float* o = out.mutable_data_ptr<float>();
+  at::parallel_for(0, S, 1, [&](int64_t sb, int64_t se) {
+    for (int64_t s = sb; s < se; ++s) {
+      float m = 0.0f;
+      for (int64_t i = ofs[s]; i < ofs[s + 1]; ++i)
+        m = std::max(m, std::fabs(x[i]));
+      o[s] = cpd::ceil_log2_abs(m * (float)world_size);
+    }
+  });
+  return out;
+}
+
+// flat[i] = Q(flat[i] * 2^shift[seg(i)])  — fused APS pre-scale + cast.
+Tensor scale_quantize_(Tensor flat, const Tensor& offsets, const Tensor& shifts,
+                       int64_t man_bits, int64_t exp_bits) {
+  check_f32_contig(flat, "flat");
+  check_f32_contig(shifts, "shifts");
+  const int64_t S = offsets.numel() - 1;
+  float* x = flat.mutable_data_ptr<float>();
+  const int64_t* ofs = offsets.const_data_ptr<int64_t>();
+  const float* sh = shifts.const_data_ptr<float>();
+  at::parallel_for(0, S, 1, [&](int64_t sb, int64_t se) {
+    for (int64_t s = sb; s < se; ++s) {
+      const float scale = std::ldexp(1.0f, (int)sh[s]);
+      for (int64_t i = ofs[s]; i < ofs[s + 1]; ++i)
+        x[i] = cpd::cast_fp(x[i] * scale, (int)man_bits, (int)exp_bits);
+    }
+  });
+  return flat;
+}
+
+// flat[i] *= 2^(sign * shift[seg(i)])  (sign=-1: APS unscale, result NOT
+// re-quantized — dist_util.py:44-45).
+Tensor seg_scale_(Tensor flat, const Tensor& offsets, const Tensor& shifts,
+                  int64_t sign) {
+  check_f32_contig(flat, "flat");
+  const int64_t S = offsets.numel() - 1;
+  float* x = flat.mutable_data_ptr<float>();
+  const int64_t* ofs = offsets.const_data_ptr<int64_t>();
+  const float* sh = shifts.const_data_ptr<float>();
+  at::parallel_for(0, S, 1, [&](int64_t sb, int64_t se) {
+    for (int64_t s = sb; s < se; ++s) {
+      const float scale = std::ldexp(1.0f, (int)(sign * sh[s]));
+      for (int64_t i = ofs[s]; i < ofs[s + 1]; ++i) x[i] *= scale;
+    }
+  });
+  return flat;
+}
+
+// C[M,N] = A[M,K] @ B[K,N] with an (exp,man)-rounded Kahan accumulator:
+// every product and every Kahan intermediate is rounded, sequentially over K
+// (reference semantics: float_kernel.cu:181-195; the reference GEMM always
+// Kahan-accumulates — the plain accumulate is commented out there).
+Tensor quant_gemm(const Tensor& a, const Tensor& b, int64_t man_bits,
+                  int64_t exp_bits) {
+  check_f32_contig(a, "a");
+  check_f32_contig(b, "b");
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0),
+              "quant_gemm: bad shapes");
+  const int64_t M = a.size(0), K = a.size(1), N = b.size(1);
+  Tensor c = at::zeros({M, N}, a.options());
+  const float* A = a.const_data_ptr<float>();
+  const float* B = b.const_data_ptr<float>();
+  float* C = c.mutable_data_ptr<float>();
+  const int mb = (int)man_bits, eb = (int)exp_bits;
+  at::parallel_for(0, M, 1, [&](int64_t rb, int64_t re) {
+    for (int64_t i = rb; i < re; ++i) {
+      for (int64_t j = 0; j < N; ++j) {
+        float acc = 0.0f, comp = 0.0f;
+        for (int64_t k = 0; k < K; ++k) {
+          const float prod = cpd::cast_fp(A[i * K + k] * B[k * N + j], mb, eb);
+          cpd::kahan_qstep(acc, comp, prod, mb, eb);
+        }
+        C[i * N + j] = acc;
+      }
+    }
+  });
+  return c;
+}
+
+// Elementwise exact ceil(log2|x|) with the -100 all-zero sentinel (small
+// helper for tests and the per-parameter APS path).
+Tensor ceil_log2(const Tensor& x) {
+  check_f32_contig(x, "x");
+  Tensor out = at::empty_like(x);
+  const float* src = x.const_data_ptr<float>();
+  float* dst = out.mutable_data_ptr<float>();
+  at::parallel_for(0, x.numel(), kGrain, [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e; ++i) dst[i] = cpd::ceil_log2_abs(src[i]);
+  });
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("quantize", &quantize, "FP32 -> (exp,man) grid, out-of-place");
+  m.def("quantize_", &quantize_, "FP32 -> (exp,man) grid, in-place");
+  m.def("qadd_", &qadd_, "acc = Q(acc + inc)");
+  m.def("kahan_qadd_", &kahan_qadd_, "quantized Kahan accumulate step");
+  m.def("seg_max_exp", &seg_max_exp, "per-segment APS max exponent");
+  m.def("scale_quantize_", &scale_quantize_, "fused per-segment scale+cast");
+  m.def("seg_scale_", &seg_scale_, "per-segment power-of-two scale");
+  m.def("quant_gemm", &quant_gemm, "GEMM with (exp,man) Kahan accumulator");
+  m.def("ceil_log2", &ceil_log2, "exact ceil(log2|x|), -100 at zero");
+}

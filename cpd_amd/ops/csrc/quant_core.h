@@ -1,0 +1,134 @@
+// cpd_amd core numerics: round an IEEE-754 binary32 value onto the grid of a
+// customized floating-point format with `exp_bits` exponent bits (<= 8) and
+// `man_bits` mantissa bits (<= 23, not counting the implicit bit).
+//
+// This single header is compiled both as host C++ (CPU ops) and as HIP device
+// code (gfx950 kernels) so the two paths are bit-identical by construction.
+//
+// Semantics (behavior-parity with the reference emulator's cast_precision,
+// /root/reference/CPDtorch/quant/quant_cuda/float_kernel.cu:10-92 — independent
+// implementation, written fresh for CDNA4):
+//   * +/-0, +/-Inf and NaN pass through unchanged (sign of zero preserved).
+//   * FP32 subnormal inputs flush to +0.
+//   * Target bias = 2^(exp_bits-1) - 1; the exponent-field value
+//     (1<<exp_bits)-1 is reserved IEEE-style, so values whose unbiased
+//     exponent reaches it saturate to +/-Inf *before* rounding.  (This is NOT
+//     OCP-FP8 e4m3fn, which has no infinities.)
+//   * Round-to-nearest-even on the mantissa at man_bits.
+//   * Because the overflow check happens before rounding, a value just below
+//     the saturation boundary whose mantissa rounds up crosses the boundary
+//     and yields the *finite* value 2^(E+1) (e.g. e4m3: 255.9 -> 256.0, not
+//     Inf).  Faithfully kept: emulation-mode and ring-mode reductions must
+//     agree bit-for-bit.
+//   * Values below the target's normal range are rounded as target-format
+//     subnormals: the mantissa (with explicit leading 1) is right-shifted
+//     first — the shifted-out sticky bits are DISCARDED before rounding, same
+//     as the reference — then rounded at man_bits.
+//   * Deliberate divergence (documented): man_bits == 23 in the subnormal
+//     path is well-defined here (no rounding) where the reference shifts by
+//     -1 (UB); and shifts >= 32 are an explicit flush-to-zero instead of
+//     hardware-dependent shift-count wrapping.
+//
+// Value reconstruction is ldexpf on the rounded integer mantissa (exact; the
+// reference's multiply-loop is O(|exponent|) and was replaced, see
+// SURVEY.md §2.1 N1).
+#pragma once
+
+#include <cstdint>
+#include <cmath>
+
+#if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
+#define CPD_HD __host__ __device__ __forceinline__
+#else
+#define CPD_HD inline
+#endif
+
+namespace cpd {
+
+CPD_HD uint32_t f32_bits(float f) {
+  uint32_t u;
+  __builtin_memcpy(&u, &f, 4);
+  return u;
+}
+
+CPD_HD float bits_f32(uint32_t u) {
+  float f;
+  __builtin_memcpy(&f, &u, 4);
+  return f;
+}
+
+// Round-to-nearest-even of a 24-bit integer significand at `man_bits`
+// fractional bits kept below the implicit-one position (bit 23).
+CPD_HD uint32_t round_mantissa_rne(uint32_t man, int man_bits) {
+  if (man_bits >= 23) return man;
+  const uint32_t drop = 23 - man_bits;          // low bits to clear
+  const uint32_t unit = 1u << drop;             // value of the kept LSB
+  const uint32_t half = unit >> 1;
+  const uint32_t rem = man & (unit - 1);
+  const bool up = (rem > half) || (rem == half && (man & unit));
+  return (man & ~(unit - 1)) + (up ? unit : 0);
+}
+
+CPD_HD float cast_fp(float x, int man_bits, int exp_bits) {
+  const uint32_t u = f32_bits(x);
+  const uint32_t exp_f = (u >> 23) & 0xFFu;
+  uint32_t man = u & 0x7FFFFFu;
+
+  if (exp_f == 0xFFu) return x;                   // Inf / NaN
+  if (exp_f == 0u) return (man == 0u) ? x : 0.0f; // +-0 kept; fp32 subnormal -> +0
+
+  const int true_exp = (int)exp_f - 127;
+  const int bias = (1 << (exp_bits - 1)) - 1;
+  const int new_e = true_exp + bias;
+  if (new_e >= (1 << exp_bits) - 1) {             // saturate to +-Inf (pre-round)
+    return bits_f32(0x7F800000u | (u & 0x80000000u));
+  }
+
+  man |= (1u << 23);                              // explicit leading one
+  int out_e;                                      // exponent of man * 2^-23
+  if (new_e > 0) {                                // target-normal
+    man = round_mantissa_rne(man, man_bits);
+    out_e = true_exp;
+  } else {                                        // target-subnormal
+    const int shift = 1 - new_e;
+    man = (shift > 31) ? 0u : (man >> shift);     // sticky discarded pre-round
+    man = round_mantissa_rne(man, man_bits);
+    out_e = 1 - bias;
+  }
+  const float mag = ldexpf((float)man, out_e - 23);
+  return (u & 0x80000000u) ? -mag : mag;
+}
+
+// One step of (exp,man)-rounded Kahan compensated accumulation:
+//   y = Q(inc - c); t = Q(acc + y); c = Q(Q(t - acc) - y); acc = t
+// Every intermediate is rounded, matching the reference's gradient-sum and
+// GEMM-accumulator semantics (float_kernel.cu:181-195, dist_util.py:82-88).
+CPD_HD void kahan_qstep(float& acc, float& c, float inc, int man_bits,
+                        int exp_bits) {
+  const float y = cast_fp(inc - c, man_bits, exp_bits);
+  const float t = cast_fp(acc + y, man_bits, exp_bits);
+  c = cast_fp(cast_fp(t - acc, man_bits, exp_bits) - y, man_bits, exp_bits);
+  acc = t;
+}
+
+// ceil(log2(|x|)) of the *magnitude* as used by APS max-exponent scanning.
+// Exact for every finite nonzero float (integer bit math, no log calls):
+//   |x| = m * 2^(e-150) with integer m in [2^23, 2^24) for normals.
+//   ceil(log2) = (e-127) when mantissa bits are all zero (exact power of 2),
+//   else (e-126).
+CPD_HD float ceil_log2_abs(float x) {
+  const uint32_t u = f32_bits(x) & 0x7FFFFFFFu;
+  if (u == 0) return -100.0f;  // all-zero sentinel (mix.py:260 uses max(...,-100))
+  const int e = (int)(u >> 23);
+  const uint32_t m = u & 0x7FFFFFu;
+  if (e == 0) {
+    // fp32 subnormal: |x| = m * 2^-149, log2 = log2(m) - 149
+    const int top = 31 - __builtin_clz(m);
+    const bool pow2 = (m & (m - 1)) == 0;
+    return (float)(top - 149 + (pow2 ? 0 : 1));
+  }
+  if (e == 0xFF) return 129.0f;  // Inf/NaN: larger than any finite exponent
+  return (float)(e - 127 + (m != 0 ? 1 : 0));
+}
+
+}  // namespace cpd

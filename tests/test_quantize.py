@@ -1,0 +1,105 @@
+"""Unit tests for the FP32 -> (exp,man) cast against the independent numpy
+oracle, on CPU (the HIP kernel is tested bit-identical in test_gpu_numerics)."""
+import numpy as np
+import pytest
+import torch
+
+from cpd_amd.quant import float_quantize, float_quantize_
+from cpd_amd.quant._oracle import cast_fp_oracle, ceil_log2_oracle
+from cpd_amd import ops
+
+FORMATS = [(4, 3), (5, 2), (8, 23), (5, 10), (8, 7), (3, 0), (2, 1), (6, 9),
+           (8, 0), (1, 0), (7, 15)]
+
+SPECIALS = np.array(
+    [0.0, -0.0, np.inf, -np.inf, np.nan, 1e-45, -1e-45, 1e-38, -1e-38,
+     240.0, 248.0, 255.9, 256.0, 448.0, -255.9, 2 ** -10, 2 ** -9,
+     0.0625, 1.0 + 2 ** -4, 1.0 + 2 ** -3, 3.4e38, -3.4e38, 65504.0,
+     2 ** -126, 2 ** -127, 2 ** -149],
+    dtype=np.float32,
+)
+
+
+def _random_bits(n, seed=0):
+    rng = np.random.default_rng(seed)
+    return rng.integers(0, 2 ** 32, size=n, dtype=np.uint32).view(np.float32)
+
+
+@pytest.mark.parametrize("exp,man", FORMATS)
+def test_cast_matches_oracle_bitexact(exp, man):
+    x = np.concatenate([_random_bits(100_000, seed=exp * 31 + man), SPECIALS])
+    got = float_quantize(torch.from_numpy(x.copy()), exp, man).numpy()
+    want = cast_fp_oracle(x, man, exp)
+    nan = np.isnan(got) & np.isnan(want)
+    assert (got.view(np.uint32) == want.view(np.uint32))[~nan].all()
+
+
+def test_special_values_e4m3():
+    f = lambda v: float_quantize(torch.tensor([v], dtype=torch.float32), 4, 3).item()
+    assert f(256.0) == np.inf          # saturates IEEE-style (not OCP e4m3fn)
+    assert f(-256.0) == -np.inf
+    assert f(255.9) == 256.0           # pre-round overflow check quirk
+    assert f(240.0) == 240.0
+    assert f(247.9) == 240.0           # RNE down: 1.936 < tie 1.9375
+    assert f(1e-45) == 0.0             # fp32 subnormal flushes
+    assert np.isnan(f(np.nan))
+    z = float_quantize(torch.tensor([-0.0]), 4, 3)
+    assert z.item() == 0.0 and np.signbit(z.numpy()[0])  # -0 preserved
+    # e4m3 subnormals: min normal 2^-6, subnormal step 2^-9
+    assert f(2 ** -9) == 2 ** -9
+    assert f(2 ** -10) == 0.0          # exact tie between 0 and 2^-9 -> even -> 0
+    assert f(1.5 * 2 ** -10) == 2 ** -9
+    assert f(2 ** -11) == 0.0
+
+
+def test_idempotent_on_grid():
+    x = torch.from_numpy(_random_bits(50_000, seed=7).copy())
+    for exp, man in [(4, 3), (5, 2), (6, 9)]:
+        q1 = float_quantize(x, exp, man)
+        finite = torch.isfinite(q1) & (q1.abs() < 2 ** (2 ** (exp - 1)))
+        q2 = float_quantize(q1, exp, man)
+        # On-grid finite values (below the overflow quirk region) are fixed points
+        assert torch.equal(q1[finite], q2[finite])
+
+
+def test_out_of_place_contract():
+    x = torch.randn(100)
+    x0 = x.clone()
+    y = float_quantize(x, 4, 3)
+    assert torch.equal(x, x0), "float_quantize must not mutate its input"
+    assert not torch.equal(y, x)
+    z = float_quantize_(x, 4, 3)
+    assert z.data_ptr() == x.data_ptr(), "float_quantize_ is in-place"
+    assert torch.equal(x, y)
+
+
+def test_ceil_log2_exact():
+    x = np.concatenate([
+        _random_bits(50_000, seed=3),
+        np.array([0.0, 1.0, 2.0, 4.0, 0.5, 3.0, 2 ** 20, 2 ** -20,
+                  np.nextafter(np.float32(2.0), np.float32(3.0)),
+                  np.nextafter(np.float32(2.0), np.float32(1.0)),
+                  2 ** -149, 2 ** -148], dtype=np.float32),
+    ])
+    x = x[np.isfinite(x)]
+    got = ops.ceil_log2(torch.from_numpy(x.copy())).numpy()
+    want = ceil_log2_oracle(x)
+    assert (got == want).all()
+    assert ops.ceil_log2(torch.zeros(1)).item() == -100.0
+
+
+def test_quantizer_autograd():
+    from cpd_amd.quant import quantizer
+
+    q = quantizer(forward_exp=4, forward_man=3, backward_exp=5, backward_man=2)
+    x = torch.randn(64, requires_grad=True)
+    y = q(x)
+    assert torch.equal(y.detach(), float_quantize(x.detach(), 4, 3))
+    g = torch.randn(64)
+    y.backward(g)
+    assert torch.equal(x.grad, float_quantize(g, 5, 2))
+
+    # (8,23) short-circuit is identity (no subnormal flush) inside quantizer
+    q_id = quantizer()
+    x2 = torch.tensor([1e-45, 1.0], requires_grad=True)
+    assert torch.equal(q_id(x2).detach(), x2.detach())
