@@ -1,0 +1,240 @@
+// cpd_amd GEMM kernels for gfx950 (CDNA4 / MI355X).
+//
+// Two hand-written kernels (no BLAS, no hipify — MI355X-first designs):
+//
+// 1. gemm_f32: general C[M,N] = A[M,K] @ B[K,N] in exact fp32 on the MFMA
+//    matrix cores (`v_mfma_f32_32x32x2_f32`).  gfx950 has no TF32/xf32 path;
+//    the f32-input MFMA is bitwise an fmaf chain at the 157 TF f32 vector
+//    rate, ~2.4x an f32 VALU GEMM (cdna_hip_programming.md §3).  Structure:
+//    128x128x32 block tile, 4 waves, each wave a 2x2 of 32x32 MFMA tiles,
+//    A staged transposed in LDS ([BK][BM+1], bank-conflict-free write via the
+//    (4k+j+m) mapping), B staged row-major.  This is the capability-parity
+//    replacement for the reference's "high performance general GEMM"
+//    (README.md:13-16) — the reference's tvm_gemm computes a 16x16 C tile per
+//    block with scalar math (float_kernel.cu:103-340).
+//
+// 2. quant_gemm: C = A @ B with an (exp,man)-rounded Kahan accumulator —
+//    every product and every Kahan intermediate is cast to the custom grid,
+//    sequentially over K (reference semantics float_kernel.cu:181-195).  The
+//    rounding between accumulation steps is inherently serial per output
+//    element, so MFMA cannot be used (its internal k-accumulation cannot be
+//    rounded); this is a VALU kernel: 64x64 C tile per 256-thread block, 4x4
+//    micro-tile per lane, LDS-staged operands.  Unlike the reference, the
+//    boundary path zero-initializes the compensation term and is barrier-
+//    correct (the reference's edge branch reads uninitialized Kahan state and
+//    lacks __syncthreads — SURVEY.md §2.1 N3; both bugs fixed by design).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include "quant_core.h"
+
+namespace {
+
+using namespace cpd;
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+// ---------------------------------------------------------------------------
+// fp32 MFMA GEMM
+// ---------------------------------------------------------------------------
+
+constexpr int BM = 128, BN = 128, BK = 32;
+
+__global__ __launch_bounds__(256) void gemm_f32_kernel(
+    const float* __restrict__ A, const float* __restrict__ B,
+    float* __restrict__ C, int M, int N, int K) {
+  __shared__ float As[BK][BM + 1];  // +1: conflict-free transposed staging
+  __shared__ float Bs[BK][BN];
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;     // 4 waves: 2x2 of 64x64 wave tiles
+  const int wr = (wave >> 1) * 64;       // wave row offset in block tile
+  const int wc = (wave & 1) * 64;
+
+  const int block_row = blockIdx.x * BM;
+  const int block_col = blockIdx.y * BN;
+
+  f32x16 acc[2][2] = {};
+
+  const int ktiles = (K + BK - 1) / BK;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    const int k0 = kt * BK;
+    // --- stage A (transposed): thread t loads A[m][4k4..4k4+3] as float4 ---
+    {
+      const int k4 = threadIdx.x & 7;          // 8 float4 per 32-wide K row
+      const int m0 = threadIdx.x >> 3;         // 32 rows per pass
+      for (int p = 0; p < 4; ++p) {
+        const int m = m0 + p * 32;
+        const int gm = block_row + m;
+        const int gk = k0 + k4 * 4;
+        float4 v = {0.f, 0.f, 0.f, 0.f};
+        if (gm < M) {
+          if (gk + 3 < K) {
+            v = *reinterpret_cast<const float4*>(A + (long)gm * K + gk);
+          } else {
+            const float* row = A + (long)gm * K;
+            if (gk + 0 < K) v.x = row[gk + 0];
+            if (gk + 1 < K) v.y = row[gk + 1];
+            if (gk + 2 < K) v.z = row[gk + 2];
+            if (gk + 3 < K) v.w = row[gk + 3];
+          }
+        }
+        As[k4 * 4 + 0][m] = v.x;
+        As[k4 * 4 + 1][m] = v.y;
+        As[k4 * 4 + 2][m] = v.z;
+        As[k4 * 4 + 3][m] = v.w;
+      }
+    }
+    // --- stage B row-major: thread t loads B[k][4n4..] as float4 ---
+    {
+      const int n4 = threadIdx.x & 31;         // 32 float4 per 128-wide row
+      const int kk0 = threadIdx.x >> 5;        // 8 k rows per pass
+      for (int p = 0; p < 4; ++p) {
+        const int kk = kk0 + p * 8;
+        const int gk = k0 + kk;
+        const int gn = block_col + n4 * 4;
+        float4 v = {0.f, 0.f, 0.f, 0.f};
+        if (gk < K) {
+          if (gn + 3 < N) {
+            v = *reinterpret_cast<const float4*>(B + (long)gk * N + gn);
+          } else {
+            const float* row = B + (long)gk * N;
+            if (gn + 0 < N) v.x = row[gn + 0];
+            if (gn + 1 < N) v.y = row[gn + 1];
+            if (gn + 2 < N) v.z = row[gn + 2];
+            if (gn + 3 < N) v.w = row[gn + 3];
+          }
+        }
+        *reinterpret_cast<float4*>(&Bs[kk][n4 * 4]) = v;
+      }
+    }
+    __syncthreads();
+
+    // --- MFMA inner loop: K advances 2 per mfma_f32_32x32x2_f32 ---
+    const int l31 = lane & 31;
+    const int khalf = lane >> 5;  // this lane's k within the 2-wide step
+    for (int kk = 0; kk < BK; kk += 2) {
+      const float a0 = As[kk + khalf][wr + l31];
+      const float a1 = As[kk + khalf][wr + 32 + l31];
+      const float b0 = Bs[kk + khalf][wc + l31];
+      const float b1 = Bs[kk + khalf][wc + 32 + l31];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // --- epilogue: C/D layout col=lane&31, row=(r&3)+8*(r>>2)+4*(lane>>5) ---
+  for (int mi = 0; mi < 2; ++mi) {
+    for (int nj = 0; nj < 2; ++nj) {
+      const int col = block_col + wc + nj * 32 + (lane & 31);
+      if (col >= N) continue;
+      for (int r = 0; r < 16; ++r) {
+        const int row = block_row + wr + mi * 32 + (r & 3) + 8 * (r >> 2) +
+                        4 * (lane >> 5);
+        if (row < M) C[(long)row * N + col] = acc[mi][nj][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// quantized-Kahan-accumulator GEMM (VALU; rounding forbids MFMA accumulation)
+// ---------------------------------------------------------------------------
+
+constexpr int QBM = 64, QBN = 64, QBK = 16;
+
+__global__ __launch_bounds__(256) void quant_gemm_kernel(
+    const float* __restrict__ A, const float* __restrict__ B,
+    float* __restrict__ C, int M, int N, int K, int man, int exp) {
+  __shared__ float As[QBK][QBM + 1];
+  __shared__ float Bs[QBK][QBN];
+
+  const int tx = threadIdx.x & 15;   // 16x16 threads, 4x4 outputs each
+  const int ty = threadIdx.x >> 4;
+  const int row0 = blockIdx.x * QBM + ty * 4;
+  const int col0 = blockIdx.y * QBN + tx * 4;
+
+  float acc[4][4] = {};
+  float comp[4][4] = {};
+
+  const int ktiles = (K + QBK - 1) / QBK;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    const int k0 = kt * QBK;
+    // stage A[64][16] transposed, B[16][64]; 256 threads x 4 elements each
+    {
+      const int k = threadIdx.x & 15;
+      const int m0 = threadIdx.x >> 4;
+      for (int p = 0; p < 4; ++p) {
+        const int m = m0 + p * 16;
+        const int gm = blockIdx.x * QBM + m;
+        As[k][m] = (gm < M && k0 + k < K) ? A[(long)gm * K + k0 + k] : 0.0f;
+      }
+      const int n = threadIdx.x & 63;
+      const int kk0 = threadIdx.x >> 6;
+      for (int p = 0; p < 4; ++p) {
+        const int kk = kk0 + p * 4;
+        const int gn = blockIdx.y * QBN + n;
+        Bs[kk][n] = (k0 + kk < K && gn < N) ? B[(long)(k0 + kk) * N + gn] : 0.0f;
+      }
+    }
+    __syncthreads();
+
+    const int klim = min(QBK, K - k0);  // never round in padded-k steps
+    for (int kk = 0; kk < klim; ++kk) {  // strictly k-ordered (semantics)
+      float a[4], b[4];
+      for (int i = 0; i < 4; ++i) a[i] = As[kk][ty * 4 + i];
+      for (int j = 0; j < 4; ++j) b[j] = Bs[kk][tx * 4 + j];
+      for (int i = 0; i < 4; ++i)
+        for (int j = 0; j < 4; ++j) {
+          const float prod = cast_fp(a[i] * b[j], man, exp);
+          kahan_qstep(acc[i][j], comp[i][j], prod, man, exp);
+        }
+    }
+    __syncthreads();
+  }
+
+  for (int i = 0; i < 4; ++i) {
+    if (row0 + i >= M) break;
+    for (int j = 0; j < 4; ++j)
+      if (col0 + j < N) C[(long)(row0 + i) * N + col0 + j] = acc[i][j];
+  }
+}
+
+inline hipStream_t cur_stream(const at::Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.get_device()).stream();
+}
+
+void check_gemm_args(const at::Tensor& a, const at::Tensor& b) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda(), "GEMM inputs must be on GPU");
+  TORCH_CHECK(a.scalar_type() == at::kFloat && b.scalar_type() == at::kFloat);
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0));
+}
+
+}  // namespace
+
+at::Tensor cpd_gemm_f32_hip(const at::Tensor& a, const at::Tensor& b) {
+  check_gemm_args(a, b);
+  const int M = a.size(0), K = a.size(1), N = b.size(1);
+  at::Tensor c = at::empty({M, N}, a.options());
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+  hipLaunchKernelGGL(gemm_f32_kernel, grid, dim3(256), 0, cur_stream(a),
+                     a.data_ptr<float>(), b.data_ptr<float>(),
+                     c.data_ptr<float>(), M, N, K);
+  return c;
+}
+
+at::Tensor cpd_quant_gemm_hip(const at::Tensor& a, const at::Tensor& b,
+                              int64_t man, int64_t exp) {
+  check_gemm_args(a, b);
+  const int M = a.size(0), K = a.size(1), N = b.size(1);
+  at::Tensor c = at::empty({M, N}, a.options());
+  dim3 grid((M + QBM - 1) / QBM, (N + QBN - 1) / QBN);
+  hipLaunchKernelGGL(quant_gemm_kernel, grid, dim3(256), 0, cur_stream(a),
+                     a.data_ptr<float>(), b.data_ptr<float>(),
+                     c.data_ptr<float>(), M, N, K, (int)man, (int)exp);
+  return c;
+}
