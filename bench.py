@@ -1,0 +1,157 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet18/CIFAR e4m3-gradient customized-precision DDP.
+
+Measures the BASELINE.json north-star metric — img/s (whole job) for ResNet18
+training with e4m3 (exp=4, man=3) gradients + APS over the real RCCL ring
+all-reduce — on synthetic CIFAR-shaped data with random-init weights, fp32
+compute (the reference's flagship computes in fp32 with fp32 master weights;
+only gradient summation is low-precision).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+N>1 is launched by the driver via torch.distributed.run (one rank per GPU,
+RCCL); rank/world/master are read from the environment.  Rank 0 prints one
+JSON line.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=40)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=512,
+                   help="per-GPU batch (reference flagship: 512, README.md:70)")
+    p.add_argument("--emulate-node", "--emulate_node", type=int, default=1,
+                   dest="emulate_node")
+    p.add_argument("--grad-exp", type=int, default=4)
+    p.add_argument("--grad-man", type=int, default=3)
+    p.add_argument("--no-aps", action="store_true")
+    p.add_argument("--use-kahan", action="store_true")
+    p.add_argument("--mode", choices=["ring", "sequential"], default="ring")
+    p.add_argument("--model", default="resnet18_cifar")
+    p.add_argument("--lr", type=float, default=0.1)
+    p.add_argument("--device", default=None)
+    p.add_argument("--channels-last", action="store_true", default=False)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    use_gpu = torch.cuda.is_available() and args.device != "cpu"
+    device = torch.device(args.device or ("cuda" if use_gpu else "cpu"))
+
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        local = int(os.environ.get("LOCAL_RANK", rank))
+        if use_gpu:
+            torch.cuda.set_device(local)
+            device = torch.device("cuda", local)
+        dist.init_process_group("nccl" if use_gpu else "gloo")
+
+    from cpd_amd.models import build_model
+    from cpd_amd.parallel import DistModule
+    from cpd_amd.trainers.core import LPTrainStep
+
+    torch.manual_seed(1234)
+    shapes = {
+        "resnet18_cifar": ((3, 32, 32), 10),
+        "resnet50": ((3, 224, 224), 1000),
+    }
+    shape, ncls = shapes[args.model]
+    model = build_model(args.model).to(device)
+    model.train()
+    dm = DistModule(model)
+    opt = torch.optim.SGD([{"params": model.parameters()}], lr=args.lr,
+                          momentum=0.9, weight_decay=1e-4)
+    step = LPTrainStep(dm, opt, grad_exp=args.grad_exp, grad_man=args.grad_man,
+                       use_APS=not args.no_aps, use_kahan=args.use_kahan,
+                       emulate_node=args.emulate_node, mode=args.mode)
+
+    # synthetic data: a small pool of fixed random batches resident on device
+    g = torch.Generator().manual_seed(42 + rank)
+    pool = [(torch.randn((args.batch,) + shape, generator=g).to(device),
+             torch.randint(0, ncls, (args.batch,), generator=g).to(device))
+            for _ in range(4)]
+    criterion = torch.nn.CrossEntropyLoss().to(device)
+    denom = step.loss_scale_denom()
+
+    def one_step(i):
+        for mb in range(args.emulate_node):
+            x, y = pool[(i * args.emulate_node + mb) % len(pool)]
+            loss = criterion(model(x), y) / denom
+            step.substep(loss)
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize()
+        if world > 1:
+            dist.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        one_step(i)
+    sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(args.warmup + i)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks -> whole-job time
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    images = args.steps * args.batch * args.emulate_node * world
+    if rank == 0:
+        result = {
+            "metric": "img/s ResNet18 e4m3-grad DDP" if
+                      args.model == "resnet18_cifar" else
+                      f"img/s {args.model} e{args.grad_exp}m{args.grad_man}-grad DDP",
+            "value": images / elapsed,
+            "unit": "img/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world * args.emulate_node,
+                "batch_per_gpu": args.batch,
+                "seq_len": None,
+                "parallelism": f"dp{world}",
+                "grad_format": f"e{args.grad_exp}m{args.grad_man}",
+                "use_APS": not args.no_aps,
+                "use_kahan": args.use_kahan,
+                "emulate_node": args.emulate_node,
+                "allreduce_mode": args.mode,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
