@@ -42,7 +42,10 @@ def parse_args():
     p.add_argument("--model", default="resnet18_cifar")
     p.add_argument("--lr", type=float, default=0.1)
     p.add_argument("--device", default=None)
-    p.add_argument("--channels-last", action="store_true", default=False)
+    p.add_argument("--channels-last", dest="channels_last",
+                   action=argparse.BooleanOptionalAction, default=True,
+                   help="NHWC memory format (MIOpen's native conv layout; "
+                        "avoids per-conv batched_transpose kernels)")
     return p.parse_args()
 
 
@@ -72,6 +75,8 @@ def main():
     }
     shape, ncls = shapes[args.model]
     model = build_model(args.model).to(device)
+    if args.channels_last and use_gpu:
+        model = model.to(memory_format=torch.channels_last)
     model.train()
     dm = DistModule(model)
     opt = torch.optim.SGD([{"params": model.parameters()}], lr=args.lr,
@@ -82,9 +87,13 @@ def main():
 
     # synthetic data: a small pool of fixed random batches resident on device
     g = torch.Generator().manual_seed(42 + rank)
-    pool = [(torch.randn((args.batch,) + shape, generator=g).to(device),
-             torch.randint(0, ncls, (args.batch,), generator=g).to(device))
-            for _ in range(4)]
+    pool = []
+    for _ in range(4):
+        x = torch.randn((args.batch,) + shape, generator=g).to(device)
+        if args.channels_last and use_gpu:
+            x = x.to(memory_format=torch.channels_last)
+        y = torch.randint(0, ncls, (args.batch,), generator=g).to(device)
+        pool.append((x, y))
     criterion = torch.nn.CrossEntropyLoss().to(device)
     denom = step.loss_scale_denom()
 

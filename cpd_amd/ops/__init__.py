@@ -91,16 +91,25 @@ def kahan_qadd_(acc, comp, inc, man_bits, exp_bits):
     return ext_for(acc).kahan_qadd_(acc, comp, inc.contiguous(), man_bits, exp_bits)
 
 
-def seg_max_exp(flat, offsets, world_size):
-    return ext_for(flat).seg_max_exp(flat, offsets, world_size)
+def seg_max_exp(flat, offsets, world_size, aligned=False):
+    # `aligned=True` (every segment boundary 256-element aligned, n%256==0,
+    # guaranteed by GradBucket) selects the wave-uniform vectorized GPU path.
+    if flat.is_cuda:
+        return hip_ext().seg_max_exp(flat, offsets, world_size, aligned)
+    return cpu_ext().seg_max_exp(flat, offsets, world_size)
 
 
-def scale_quantize_(flat, offsets, shifts, man_bits, exp_bits):
-    return ext_for(flat).scale_quantize_(flat, offsets, shifts, man_bits, exp_bits)
+def scale_quantize_(flat, offsets, shifts, man_bits, exp_bits, aligned=False):
+    if flat.is_cuda:
+        return hip_ext().scale_quantize_(flat, offsets, shifts, man_bits,
+                                         exp_bits, aligned)
+    return cpu_ext().scale_quantize_(flat, offsets, shifts, man_bits, exp_bits)
 
 
-def seg_scale_(flat, offsets, shifts, sign):
-    return ext_for(flat).seg_scale_(flat, offsets, shifts, sign)
+def seg_scale_(flat, offsets, shifts, sign, aligned=False):
+    if flat.is_cuda:
+        return hip_ext().seg_scale_(flat, offsets, shifts, sign, aligned)
+    return cpu_ext().seg_scale_(flat, offsets, shifts, sign)
 
 
 def quant_gemm_raw(a, b, man_bits, exp_bits):

@@ -32,10 +32,22 @@ class GradBucket:
                                     device=device)
         self.attach()
 
+    @staticmethod
+    def _grad_view(flat_slice, p):
+        # Match the parameter's memory format so autograd accumulates into
+        # the bucket without a strided elementwise path (channels_last conv
+        # weights produce channels_last grads).
+        if (p.dim() == 4 and not p.is_contiguous()
+                and p.is_contiguous(memory_format=torch.channels_last)):
+            n, c, h, w = p.shape
+            return flat_slice.as_strided((n, c, h, w),
+                                         (c * h * w, 1, w * c, c))
+        return flat_slice.view_as(p)
+
     def attach(self):
         """(Re)point every param.grad at its view of the flat buffer."""
         for p, st in zip(self.params, self.starts):
-            view = self.flat[st:st + p.numel()].view_as(p)
+            view = self._grad_view(self.flat[st:st + p.numel()], p)
             if p.grad is None or p.grad.data_ptr() != view.data_ptr():
                 if p.grad is not None:
                     view.copy_(p.grad.detach())

@@ -173,14 +173,15 @@ def _sum_gradients_fused(bucket, use_APS, grad_exp, grad_man, use_kahan,
 
     shifts = None
     if use_APS:
-        shifts = ops.seg_max_exp(flat, offsets, W)   # [S], on device
+        shifts = ops.seg_max_exp(flat, offsets, W, aligned=True)  # [S], device
         if distributed:
             dist.all_reduce(shifts, op=dist.ReduceOp.MAX)
         upper = float(2 ** (grad_exp - 1) - 1)
         # shift = upper - E  (E = -100 sentinel for all-zero grads is safe:
         # 0 * 2^(upper+100) == 0)
         shifts = (upper - shifts).float()
-        ops.scale_quantize_(flat, offsets, shifts, grad_man, grad_exp)
+        ops.scale_quantize_(flat, offsets, shifts, grad_man, grad_exp,
+                            aligned=True)
         if wire is None and flat.is_cuda and grad_man <= 7:
             wire = "bf16"  # values are on-grid after scale_quantize_
 
@@ -192,7 +193,7 @@ def _sum_gradients_fused(bucket, use_APS, grad_exp, grad_man, use_kahan,
         lp_all_reduce_single_(flat, grad_exp, grad_man, use_kahan)
 
     if shifts is not None:
-        ops.seg_scale_(flat, offsets, shifts, -1)
+        ops.seg_scale_(flat, offsets, shifts, -1, aligned=True)
 
 
 def lp_all_reduce_single_(flat, grad_exp, grad_man, use_kahan):

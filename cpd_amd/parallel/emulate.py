@@ -45,7 +45,8 @@ class NodeEmulator:
 
         # per-segment max exponent over ALL buffered copies, world factor = N
         max_exp = torch.stack([
-            ops.seg_max_exp(b, offsets, self.n) for b in self.buffers
+            ops.seg_max_exp(b, offsets, self.n, aligned=True)
+            for b in self.buffers
         ]).amax(0).contiguous()
         upper = float(2 ** (grad_exp - 1) - 1)
         shifts = upper - max_exp
@@ -59,11 +60,12 @@ class NodeEmulator:
         res = torch.zeros_like(flat)
         comp = torch.zeros_like(flat) if use_kahan else None
         for b in self.buffers:
-            ops.scale_quantize_(b, offsets, shifts, grad_man, grad_exp)
+            ops.scale_quantize_(b, offsets, shifts, grad_man, grad_exp,
+                                aligned=True)
             if use_kahan:
                 ops.kahan_qadd_(res, comp, b, grad_man, grad_exp)
             else:
                 ops.qadd_(res, b, grad_man, grad_exp)
         flat.copy_(res)
-        ops.seg_scale_(flat, offsets, shifts, -1)
+        ops.seg_scale_(flat, offsets, shifts, -1, aligned=True)
         self.buffers.clear()

@@ -68,9 +68,12 @@ def sequential_lp_all_reduce_(flat, grad_exp, grad_man, use_kahan=False,
                               group=None):
     """Reference-emulation semantics: all-gather + in-order quantized sum
     (dist_util.py:60-89).  Bit-identical on every rank by construction."""
-    W = dist.get_world_size(group)
-    gather = [torch.empty_like(flat) for _ in range(W)]
-    dist.all_gather(gather, flat, group=group)
+    if dist.is_available() and dist.is_initialized():
+        W = dist.get_world_size(group)
+        gather = [torch.empty_like(flat) for _ in range(W)]
+        dist.all_gather(gather, flat, group=group)
+    else:
+        gather = [flat.clone()]
     res = torch.zeros_like(flat)
     if use_kahan:
         comp = torch.zeros_like(flat)
@@ -86,8 +89,11 @@ def sequential_lp_all_reduce_(flat, grad_exp, grad_man, use_kahan=False,
 def ring_lp_all_reduce_(flat, grad_exp, grad_man, use_kahan=False, group=None,
                         wire=None):
     """Real ring all-reduce with per-hop quantized partial sums."""
-    W = dist.get_world_size(group)
-    r = dist.get_rank(group)
+    if dist.is_available() and dist.is_initialized():
+        W = dist.get_world_size(group)
+        r = dist.get_rank(group)
+    else:
+        W, r = 1, 0
     man, exp = grad_man, grad_exp
     n = flat.numel()
 
