@@ -148,3 +148,36 @@ def test_native_extension_is_loaded():
     ext = ops.hip_ext()
     assert ext is not None
     import _cpd_hip  # noqa: F401  (in-tree .so on sys.path via ops)
+
+
+def test_seg_aligned_matches_generic():
+    """The wave-uniform aligned fast path must agree with the generic path
+    and with the CPU extension on a bucket-shaped (256-aligned) layout."""
+    torch.manual_seed(9)
+    sizes = [1024, 2048, 256, 51200, 7168]
+    offsets_c = torch.tensor(np.concatenate([[0], np.cumsum(sizes)]),
+                             dtype=torch.int64)
+    n = int(offsets_c[-1])
+    flat_c = torch.randn(n) * 5
+    flat_g = flat_c.cuda()
+    offsets_g = offsets_c.cuda()
+
+    me_cpu = ops.seg_max_exp(flat_c, offsets_c, 4)
+    me_a = ops.seg_max_exp(flat_g, offsets_g, 4, aligned=True)
+    me_gen = ops.seg_max_exp(flat_g, offsets_g, 4, aligned=False)
+    assert torch.equal(me_a.cpu(), me_cpu)
+    assert torch.equal(me_gen.cpu(), me_cpu)
+
+    shifts_c = torch.tensor([1.0, -2.0, 3.0, 0.0, 5.0])
+    want = flat_c.clone()
+    ops.scale_quantize_(want, offsets_c, shifts_c, 3, 4)
+    got_a = flat_g.clone()
+    ops.scale_quantize_(got_a, offsets_g, shifts_c.cuda(), 3, 4, aligned=True)
+    got_g = flat_g.clone()
+    ops.scale_quantize_(got_g, offsets_g, shifts_c.cuda(), 3, 4, aligned=False)
+    assert torch.equal(got_a.cpu(), want)
+    assert torch.equal(got_g.cpu(), want)
+
+    ops.seg_scale_(want, offsets_c, shifts_c, -1)
+    ops.seg_scale_(got_a, offsets_g, shifts_c.cuda(), -1, aligned=True)
+    assert torch.equal(got_a.cpu(), want)
