@@ -43,9 +43,10 @@ def parse_args():
     p.add_argument("--lr", type=float, default=0.1)
     p.add_argument("--device", default=None)
     p.add_argument("--channels-last", dest="channels_last",
-                   action=argparse.BooleanOptionalAction, default=True,
-                   help="NHWC memory format (MIOpen's native conv layout; "
-                        "avoids per-conv batched_transpose kernels)")
+                   action=argparse.BooleanOptionalAction, default=False,
+                   help="NHWC memory format.  Measured 20x SLOWER for fp32 "
+                        "CIFAR-size convs on MIOpen (falls off the tuned "
+                        "igemm path); NCHW is the fast default")
     return p.parse_args()
 
 
