@@ -115,6 +115,10 @@ class DistModule(Module):
         super().__init__()
         self.module = module
         broadcast_params(self.module)
+        # the flat bucket is fp32; low-precision models (e.g. DavidNet --half)
+        # keep per-parameter grads and go through the unfused path
+        fuse = fuse and all(p.dtype == torch.float32
+                            for p in module.parameters() if p.requires_grad)
         self.bucket = GradBucket(self.module.parameters()) if fuse else None
 
     def forward(self, *inputs, **kwargs):
@@ -246,7 +250,7 @@ def normal_sum_gradients(model, grad_exp=8, grad_man=23, mode="sequential"):
     for p in model.parameters():
         if not (p.requires_grad and p.grad is not None):
             continue
-        g = p.grad.view(-1).contiguous()
+        g = p.grad.detach().reshape(-1).contiguous().float()
         if distributed and _world() > 1:
             lp_all_reduce_(g, grad_exp, grad_man, use_kahan=False, mode=mode,
                            wire="f32")
@@ -263,7 +267,7 @@ def kahan_sum_gradients(model, grad_exp=8, grad_man=23, mode="sequential"):
     for p in model.parameters():
         if not (p.requires_grad and p.grad is not None):
             continue
-        g = p.grad.view(-1).contiguous()
+        g = p.grad.detach().reshape(-1).contiguous().float()
         if distributed and _world() > 1:
             lp_all_reduce_(g, grad_exp, grad_man, use_kahan=True, mode=mode,
                            wire="f32")

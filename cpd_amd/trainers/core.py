@@ -23,7 +23,9 @@ class LPTrainStep:
     def __init__(self, model: DistModule, optimizer, *, grad_exp=4, grad_man=3,
                  use_APS=True, use_kahan=False, emulate_node=1, mode="ring",
                  use_master=True, distributed=None):
-        assert model.bucket is not None, "LPTrainStep needs a fused DistModule"
+        if model.bucket is None:
+            assert emulate_node == 1, \
+                "emulate_node > 1 needs the fused (fp32) bucket path"
         self.model = model
         self.optimizer = optimizer
         self.grad_exp = grad_exp
