@@ -70,6 +70,7 @@ def main():
     from cpd_amd.trainers.core import LPTrainStep
 
     torch.manual_seed(1234)
+    torch.backends.cudnn.benchmark = True  # MIOpen find: tune conv algos once
     shapes = {
         "resnet18_cifar": ((3, 32, 32), 10),
         "resnet50": ((3, 224, 224), 1000),
