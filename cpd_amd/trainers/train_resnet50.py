@@ -75,23 +75,16 @@ def main(argv=None):
     model = models.resnet50().to(device).train()
     dm = DistModule(model)
 
-    # linear-scaled LR over total effective batch (main.py:126-131)
+    # linear-scaled LR over total effective batch, BN params wd=0
+    # (main.py:123-131 behavior)
     lr_scale = args.batches_per_allreduce * world_size
     rest, bn_params = bn_param_split(model)
     optimizer = torch.optim.SGD(
-        [{'params': rest + bn_params}], lr=args.base_lr * lr_scale,
-        momentum=args.momentum, weight_decay=args.wd, nesterov=True)
-    # NOTE: wd split is applied manually below because LPTrainStep's master
-    # machinery repoints a single group; BN wd-0 is emulated by zeroing wd
-    # contribution — simplest faithful version keeps one group and relies on
-    # the reference behavior only when masters are off:
-    step = LPTrainStep(dm, optimizer, grad_exp=args.grad_exp,
-                       grad_man=args.grad_man, use_APS=args.use_APS,
-                       use_kahan=args.use_kahan,
-                       emulate_node=args.batches_per_allreduce,
-                       mode=args.mode)
-
-    # auto-resume: scan checkpoint-{epoch} downward (main.py:70-75)
+        [{'params': rest, 'weight_decay': args.wd},
+         {'params': bn_params, 'weight_decay': 0.0}],
+        lr=args.base_lr * lr_scale, momentum=args.momentum, nesterov=True)
+    # auto-resume: scan checkpoint-{epoch} downward (main.py:70-75) —
+    # BEFORE the master copies are made, so they snapshot the resumed weights
     start_epoch = 0
     for ep in range(args.epochs, 0, -1):
         path = args.checkpoint_format.format(epoch=ep)
@@ -101,6 +94,12 @@ def main(argv=None):
             optimizer.load_state_dict(ckpt['optimizer'])
             start_epoch = ep
             break
+
+    step = LPTrainStep(dm, optimizer, grad_exp=args.grad_exp,
+                       grad_man=args.grad_man, use_APS=args.use_APS,
+                       use_kahan=args.use_kahan,
+                       emulate_node=args.batches_per_allreduce,
+                       mode=args.mode)
 
     train_set = SyntheticImages(args.synthetic_size,
                                 shape=(3, args.image_size, args.image_size),
