@@ -47,6 +47,9 @@ def parse_args():
     p.add_argument("--model", default="resnet18_cifar")
     p.add_argument("--lr", type=float, default=0.1)
     p.add_argument("--device", default=None)
+    p.add_argument("--fused-bn", dest="fused_bn",
+                   action=argparse.BooleanOptionalAction, default=True,
+                   help="fused BN(+add)+ReLU gfx950 kernels")
     p.add_argument("--channels-last", dest="channels_last",
                    action=argparse.BooleanOptionalAction, default=False,
                    help="NHWC memory format.  Measured 20x SLOWER for fp32 "
@@ -81,7 +84,8 @@ def main():
         "resnet50": ((3, 224, 224), 1000),
     }
     shape, ncls = shapes[args.model]
-    model = build_model(args.model).to(device)
+    model = build_model(args.model,
+                        fused_bn=args.fused_bn and use_gpu).to(device)
     if args.channels_last and use_gpu:
         model = model.to(memory_format=torch.channels_last)
     model.train()

@@ -1,0 +1,83 @@
+"""Fused BatchNorm2d(+residual add)(+ReLU) module backed by the gfx950
+kernels in ops/csrc/bn_fused.hip.
+
+Replaces the eager chain {MIOpen BN, add, ReLU, threshold_backward} with two
+fused HBM passes forward and two backward.  State-dict compatible with
+nn.BatchNorm2d (same parameter/buffer names).  Falls back to the eager
+composition on CPU, in eval mode, or for shapes the kernels don't cover —
+on GPU training the fused path is the one that runs (fails loudly if the HIP
+extension is missing).
+"""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+
+
+class _FusedBNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var, momentum,
+                eps, relu, residual):
+        ext = ops.hip_ext()
+        y, mean, invstd = ext.bn_relu_fwd(
+            x, weight, bias, running_mean, running_var, momentum, eps, relu,
+            residual)
+        ctx.relu = relu
+        ctx.has_res = residual is not None
+        if relu:
+            ctx.save_for_backward(x, weight, mean, invstd, y)
+        else:
+            ctx.save_for_backward(x, weight, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        if ctx.relu:
+            x, weight, mean, invstd, y = ctx.saved_tensors
+        else:
+            x, weight, mean, invstd = ctx.saved_tensors
+            y = None
+        ext = ops.hip_ext()
+        dx, dgamma, dbeta, dres = ext.bn_relu_bwd(
+            x, dy, y, mean, invstd, weight, ctx.has_res)
+        return (dx, dgamma, dbeta, None, None, None, None, None,
+                dres if ctx.has_res else None)
+
+
+class FusedBNReLU(nn.Module):
+    """BatchNorm2d with optional fused ReLU and fused residual add
+    (``forward(x, residual=None)`` computes relu(bn(x) + residual))."""
+
+    def __init__(self, num_features, relu=True, eps=1e-5, momentum=0.1):
+        super().__init__()
+        self.num_features = num_features
+        self.relu = relu
+        self.eps = eps
+        self.momentum = momentum
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+        self.register_buffer("num_batches_tracked",
+                             torch.tensor(0, dtype=torch.long))
+
+    def _eager(self, x, residual):
+        y = F.batch_norm(x, self.running_mean, self.running_var, self.weight,
+                         self.bias, self.training, self.momentum, self.eps)
+        if residual is not None:
+            y = y + residual
+        return F.relu(y) if self.relu else y
+
+    def forward(self, x, residual=None):
+        use_fused = (self.training and x.is_cuda
+                     and x.dtype == torch.float32
+                     and (x.shape[2] * x.shape[3]) % 4 == 0)
+        if not use_fused:
+            return self._eager(x, residual)
+        if self.training:
+            self.num_batches_tracked += 1
+        res = residual.contiguous() if residual is not None else None
+        return _FusedBNFn.apply(x.contiguous(), self.weight, self.bias,
+                                self.running_mean, self.running_var,
+                                self.momentum, self.eps, self.relu, res)

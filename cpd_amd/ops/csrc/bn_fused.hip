@@ -1,0 +1,338 @@
+// Fused BatchNorm2d(+residual add)+ReLU training kernels for gfx950 (NCHW).
+//
+// MI355X-first fusion: the eager path runs BN (3 MIOpen kernels), a separate
+// add, a separate ReLU, and threshold_backward — all HBM-bound elementwise
+// passes.  Here forward is {deterministic two-level reduce, normalize+add+
+// relu} and backward is {reduce of (dy_eff, dy_eff*xhat), finalize, dx(+dres)}
+// with the ReLU mask and residual add folded in.  All reductions are
+// deterministic (fixed-shape workspace, no float atomics).
+//
+// Layout: NCHW fp32, HW % 4 == 0 (float4 I/O).  Channel c of sample n is a
+// contiguous run of HW floats at (n*C + c)*HW.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+namespace {
+
+constexpr int TPB = 256;
+
+__device__ __forceinline__ float4 ld4(const float* p) {
+  return *reinterpret_cast<const float4*>(p);
+}
+__device__ __forceinline__ void st4(float* p, float4 v) {
+  *reinterpret_cast<float4*>(p) = v;
+}
+
+__device__ float block_reduce(float v, float* smem) {
+  for (int d = 32; d > 0; d >>= 1) v += __shfl_down(v, d);
+  const int wave = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) smem[wave] = v;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < TPB / 64; ++w) v += smem[w];
+    smem[0] = v;
+  }
+  __syncthreads();
+  return smem[0];
+}
+
+// ---- forward pass 1: per (channel, slice) partial sum / sumsq -------------
+__global__ void bn_reduce_kernel(const float* __restrict__ x, int N, int C,
+                                 long HW, int split,
+                                 float* __restrict__ ws /* [C][split][2] */) {
+  const int c = blockIdx.x;
+  const int s = blockIdx.y;
+  __shared__ float smem[TPB / 64];
+  float sum = 0.f, sumsq = 0.f;
+  for (int n = s; n < N; n += split) {
+    const float* base = x + ((long)n * C + c) * HW;
+    for (long i = threadIdx.x * 4; i < HW; i += (long)TPB * 4) {
+      const float4 v = ld4(base + i);
+      sum += v.x + v.y + v.z + v.w;
+      sumsq += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    }
+  }
+  const float bs = block_reduce(sum, smem);
+  __syncthreads();
+  const float bq = block_reduce(sumsq, smem);
+  if (threadIdx.x == 0) {
+    ws[((long)c * split + s) * 2 + 0] = bs;
+    ws[((long)c * split + s) * 2 + 1] = bq;
+  }
+}
+
+// ---- forward pass 1b: finalize mean/invstd + update running stats ---------
+__global__ void bn_finalize_kernel(const float* __restrict__ ws, int C,
+                                   int split, float count, float eps,
+                                   float momentum,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var) {
+  const int c = blockIdx.x * TPB + threadIdx.x;
+  if (c >= C) return;
+  float sum = 0.f, sumsq = 0.f;
+  for (int s = 0; s < split; ++s) {
+    sum += ws[((long)c * split + s) * 2 + 0];
+    sumsq += ws[((long)c * split + s) * 2 + 1];
+  }
+  const float m = sum / count;
+  const float var = fmaxf(sumsq / count - m * m, 0.0f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+  if (running_mean) {
+    running_mean[c] += momentum * (m - running_mean[c]);
+    // running_var uses the unbiased estimator (torch semantics)
+    const float unbiased = var * count / fmaxf(count - 1.0f, 1.0f);
+    running_var[c] += momentum * (unbiased - running_var[c]);
+  }
+}
+
+// ---- forward pass 2: y = relu(xhat*gamma + beta [+ res]) ------------------
+__global__ void bn_norm_kernel(const float* __restrict__ x,
+                               const float* __restrict__ res,  // nullable
+                               float* __restrict__ y, int N, int C, long HW,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ invstd,
+                               const float* __restrict__ gamma,
+                               const float* __restrict__ beta, int relu) {
+  const long total4 = (long)N * C * HW / 4;
+  const long stride = (long)gridDim.x * TPB;
+  for (long i4 = (long)blockIdx.x * TPB + threadIdx.x; i4 < total4;
+       i4 += stride) {
+    const long i = i4 * 4;
+    const int c = (int)((i / HW) % C);
+    const float a = gamma[c] * invstd[c];
+    const float b = beta[c] - mean[c] * a;
+    float4 v = ld4(x + i);
+    v.x = v.x * a + b;
+    v.y = v.y * a + b;
+    v.z = v.z * a + b;
+    v.w = v.w * a + b;
+    if (res) {
+      const float4 r = ld4(res + i);
+      v.x += r.x; v.y += r.y; v.z += r.z; v.w += r.w;
+    }
+    if (relu) {
+      v.x = fmaxf(v.x, 0.f);
+      v.y = fmaxf(v.y, 0.f);
+      v.z = fmaxf(v.z, 0.f);
+      v.w = fmaxf(v.w, 0.f);
+    }
+    st4(y + i, v);
+  }
+}
+
+// ---- backward pass 1: per (c, slice) partials of sum(dy_eff),
+//      sum(dy_eff * xhat); dy_eff = dy * (y > 0) when relu ----------------
+__global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
+                                     const float* __restrict__ dy,
+                                     const float* __restrict__ y,  // nullable
+                                     int N, int C, long HW, int split,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     float* __restrict__ ws /*[C][split][2]*/) {
+  const int c = blockIdx.x;
+  const int s = blockIdx.y;
+  __shared__ float smem[TPB / 64];
+  const float m = mean[c], is = invstd[c];
+  float sd = 0.f, sdx = 0.f;
+  for (int n = s; n < N; n += split) {
+    const long off = ((long)n * C + c) * HW;
+    for (long i = threadIdx.x * 4; i < HW; i += (long)TPB * 4) {
+      float4 g = ld4(dy + off + i);
+      const float4 v = ld4(x + off + i);
+      if (y) {
+        const float4 yy = ld4(y + off + i);
+        g.x = yy.x > 0.f ? g.x : 0.f;
+        g.y = yy.y > 0.f ? g.y : 0.f;
+        g.z = yy.z > 0.f ? g.z : 0.f;
+        g.w = yy.w > 0.f ? g.w : 0.f;
+      }
+      sd += g.x + g.y + g.z + g.w;
+      sdx += g.x * (v.x - m) + g.y * (v.y - m) + g.z * (v.z - m) +
+             g.w * (v.w - m);
+    }
+  }
+  const float bs = block_reduce(sd, smem);
+  __syncthreads();
+  const float bq = block_reduce(sdx * 1.0f, smem);
+  if (threadIdx.x == 0) {
+    ws[((long)c * split + s) * 2 + 0] = bs;
+    ws[((long)c * split + s) * 2 + 1] = bq * is;  // sum(dy_eff * xhat)
+  }
+}
+
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ ws, int C,
+                                       int split,
+                                       float* __restrict__ sum_dy,
+                                       float* __restrict__ sum_dyx,
+                                       float* __restrict__ dgamma,
+                                       float* __restrict__ dbeta) {
+  const int c = blockIdx.x * TPB + threadIdx.x;
+  if (c >= C) return;
+  float sd = 0.f, sdx = 0.f;
+  for (int s = 0; s < split; ++s) {
+    sd += ws[((long)c * split + s) * 2 + 0];
+    sdx += ws[((long)c * split + s) * 2 + 1];
+  }
+  sum_dy[c] = sd;
+  sum_dyx[c] = sdx;
+  dbeta[c] = sd;
+  dgamma[c] = sdx;
+}
+
+// ---- backward pass 2: dx (+ dres = dy_eff) --------------------------------
+__global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
+                                 const float* __restrict__ dy,
+                                 const float* __restrict__ y,  // nullable
+                                 float* __restrict__ dx,
+                                 float* __restrict__ dres,  // nullable
+                                 int N, int C, long HW,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ sum_dy,
+                                 const float* __restrict__ sum_dyx,
+                                 float inv_count) {
+  const long total4 = (long)N * C * HW / 4;
+  const long stride = (long)gridDim.x * TPB;
+  for (long i4 = (long)blockIdx.x * TPB + threadIdx.x; i4 < total4;
+       i4 += stride) {
+    const long i = i4 * 4;
+    const int c = (int)((i / HW) % C);
+    const float m = mean[c], is = invstd[c];
+    const float k = gamma[c] * is;
+    const float md = sum_dy[c] * inv_count;
+    const float mdx = sum_dyx[c] * inv_count;
+    float4 g = ld4(dy + i);
+    const float4 v = ld4(x + i);
+    if (y) {
+      const float4 yy = ld4(y + i);
+      g.x = yy.x > 0.f ? g.x : 0.f;
+      g.y = yy.y > 0.f ? g.y : 0.f;
+      g.z = yy.z > 0.f ? g.z : 0.f;
+      g.w = yy.w > 0.f ? g.w : 0.f;
+    }
+    if (dres) st4(dres + i, g);
+    float4 o;
+    o.x = k * (g.x - md - (v.x - m) * is * mdx);
+    o.y = k * (g.y - md - (v.y - m) * is * mdx);
+    o.z = k * (g.z - md - (v.z - m) * is * mdx);
+    o.w = k * (g.w - md - (v.w - m) * is * mdx);
+    st4(dx + i, o);
+  }
+}
+
+// ---------------------------------------------------------------------------
+using at::Tensor;
+
+inline hipStream_t cur_stream(const Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.get_device()).stream();
+}
+
+inline int elem_grid(long total4) {
+  return (int)std::min<long>((total4 + TPB - 1) / TPB, 16384);
+}
+
+int pick_split(int N, int C, long HW) {
+  // enough blocks to fill 256 CUs a few times over
+  long per_split_blocks = C;
+  int split = 1;
+  while (per_split_blocks * split < 2048 && split < N) split *= 2;
+  return std::min(split, N);
+}
+
+std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
+                                const Tensor& beta, Tensor running_mean,
+                                Tensor running_var, double momentum,
+                                double eps, bool relu,
+                                const c10::optional<Tensor>& residual) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous() &&
+              x.scalar_type() == at::kFloat);
+  const int N = x.size(0), C = x.size(1);
+  const long HW = (long)x.size(2) * x.size(3);
+  TORCH_CHECK(HW % 4 == 0, "fused BN needs H*W % 4 == 0");
+  const int split = pick_split(N, C, HW);
+  auto opts = x.options();
+  Tensor ws = at::empty({C, split, 2}, opts);
+  Tensor mean = at::empty({C}, opts);
+  Tensor invstd = at::empty({C}, opts);
+  Tensor y = at::empty_like(x);
+  auto st = cur_stream(x);
+  hipLaunchKernelGGL(bn_reduce_kernel, dim3(C, split), dim3(TPB), 0, st,
+                     x.data_ptr<float>(), N, C, HW, split,
+                     ws.data_ptr<float>());
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + TPB - 1) / TPB), dim3(TPB),
+                     0, st, ws.data_ptr<float>(), C, split,
+                     (float)((long)N * HW), (float)eps, (float)momentum,
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     running_mean.defined() ? running_mean.data_ptr<float>()
+                                            : nullptr,
+                     running_var.defined() ? running_var.data_ptr<float>()
+                                           : nullptr);
+  const float* res_ptr = nullptr;
+  if (residual.has_value()) {
+    TORCH_CHECK(residual->is_contiguous() && residual->sizes() == x.sizes());
+    res_ptr = residual->data_ptr<float>();
+  }
+  hipLaunchKernelGGL(bn_norm_kernel, dim3(elem_grid((long)N * C * HW / 4)),
+                     dim3(TPB), 0, st, x.data_ptr<float>(), res_ptr,
+                     y.data_ptr<float>(), N, C, HW, mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                     beta.data_ptr<float>(), relu ? 1 : 0);
+  return {y, mean, invstd};
+}
+
+std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
+                                const c10::optional<Tensor>& y_for_mask,
+                                const Tensor& mean, const Tensor& invstd,
+                                const Tensor& gamma, bool need_dres) {
+  const int N = x.size(0), C = x.size(1);
+  const long HW = (long)x.size(2) * x.size(3);
+  const int split = pick_split(N, C, HW);
+  auto opts = x.options();
+  Tensor ws = at::empty({C, split, 2}, opts);
+  Tensor sum_dy = at::empty({C}, opts);
+  Tensor sum_dyx = at::empty({C}, opts);
+  Tensor dgamma = at::empty({C}, opts);
+  Tensor dbeta = at::empty({C}, opts);
+  Tensor dx = at::empty_like(x);
+  Tensor dres;
+  auto st = cur_stream(x);
+  const float* yp = y_for_mask.has_value() ? y_for_mask->data_ptr<float>()
+                                           : nullptr;
+  Tensor dyc = dy.contiguous();
+  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C, split), dim3(TPB), 0, st,
+                     x.data_ptr<float>(), dyc.data_ptr<float>(), yp, N, C, HW,
+                     split, mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     ws.data_ptr<float>());
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + TPB - 1) / TPB),
+                     dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
+                     sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
+  float* dres_ptr = nullptr;
+  if (need_dres) {
+    dres = at::empty_like(x);
+    dres_ptr = dres.data_ptr<float>();
+  }
+  hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(elem_grid((long)N * C * HW / 4)),
+                     dim3(TPB), 0, st, x.data_ptr<float>(),
+                     dyc.data_ptr<float>(), yp, dx.data_ptr<float>(), dres_ptr,
+                     N, C, HW, mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                     sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+                     1.0f / (float)((long)N * HW));
+  if (!need_dres) dres = at::Tensor();
+  return {dx, dgamma, dbeta, dres};
+}
+
+}  // namespace
+
+void cpd_register_bn(pybind11::module_& m) {
+  m.def("bn_relu_fwd", &bn_relu_fwd,
+        "fused BN(+res)+ReLU forward: returns (y, mean, invstd)");
+  m.def("bn_relu_bwd", &bn_relu_bwd,
+        "fused BN(+res)+ReLU backward: returns (dx, dgamma, dbeta, dres)");
+}

@@ -507,8 +507,10 @@ Tensor ceil_log2(const Tensor& x) {
 at::Tensor cpd_quant_gemm_hip(const at::Tensor& a, const at::Tensor& b,
                               int64_t man, int64_t exp);
 at::Tensor cpd_gemm_f32_hip(const at::Tensor& a, const at::Tensor& b);
+void cpd_register_bn(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  cpd_register_bn(m);
   m.def("quantize", &quantize);
   m.def("quantize_", &quantize_);
   m.def("qadd_", &qadd_);
