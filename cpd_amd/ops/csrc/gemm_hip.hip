@@ -40,90 +40,126 @@ using f32x16 = __attribute__((ext_vector_type(16))) float;
 
 constexpr int BM = 128, BN = 128, BK = 32;
 
+// Load one K-tile's worth of A/B for this thread into registers (guarded
+// for arbitrary M/N/K edges).  4 float4 of A + 4 float4 of B per thread.
+struct StageRegs {
+  float4 a[4];
+  float4 b[4];
+};
+
+__device__ __forceinline__ void stage_load(const float* __restrict__ A,
+                                           const float* __restrict__ B,
+                                           int M, int N, int K, int block_row,
+                                           int block_col, int k0, int tid,
+                                           StageRegs& r) {
+  const int k4 = tid & 7;                  // A: 8 float4 per 32-wide K row
+  const int m0 = tid >> 3;                 //    32 rows per pass
+  for (int p = 0; p < 4; ++p) {
+    const int gm = block_row + m0 + p * 32;
+    const int gk = k0 + k4 * 4;
+    float4 v = {0.f, 0.f, 0.f, 0.f};
+    if (gm < M) {
+      if (gk + 3 < K) {
+        v = *reinterpret_cast<const float4*>(A + (long)gm * K + gk);
+      } else {
+        const float* row = A + (long)gm * K;
+        if (gk + 0 < K) v.x = row[gk + 0];
+        if (gk + 1 < K) v.y = row[gk + 1];
+        if (gk + 2 < K) v.z = row[gk + 2];
+      }
+    }
+    r.a[p] = v;
+  }
+  const int n4 = tid & 31;                 // B: 32 float4 per 128-wide row
+  const int kk0 = tid >> 5;                //    8 k rows per pass
+  for (int p = 0; p < 4; ++p) {
+    const int gk = k0 + kk0 + p * 8;
+    const int gn = block_col + n4 * 4;
+    float4 v = {0.f, 0.f, 0.f, 0.f};
+    if (gk < K) {
+      if (gn + 3 < N) {
+        v = *reinterpret_cast<const float4*>(B + (long)gk * N + gn);
+      } else {
+        const float* row = B + (long)gk * N;
+        if (gn + 0 < N) v.x = row[gn + 0];
+        if (gn + 1 < N) v.y = row[gn + 1];
+        if (gn + 2 < N) v.z = row[gn + 2];
+      }
+    }
+    r.b[p] = v;
+  }
+}
+
+__device__ __forceinline__ void stage_write(float (*As)[BM + 1],
+                                            float (*Bs)[BN], int tid,
+                                            const StageRegs& r) {
+  const int k4 = tid & 7;
+  const int m0 = tid >> 3;
+  for (int p = 0; p < 4; ++p) {
+    const int m = m0 + p * 32;
+    As[k4 * 4 + 0][m] = r.a[p].x;
+    As[k4 * 4 + 1][m] = r.a[p].y;
+    As[k4 * 4 + 2][m] = r.a[p].z;
+    As[k4 * 4 + 3][m] = r.a[p].w;
+  }
+  const int n4 = tid & 31;
+  const int kk0 = tid >> 5;
+  for (int p = 0; p < 4; ++p)
+    *reinterpret_cast<float4*>(&Bs[kk0 + p * 8][n4 * 4]) = r.b[p];
+}
+
 __global__ __launch_bounds__(256) void gemm_f32_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     float* __restrict__ C, int M, int N, int K) {
-  __shared__ float As[BK][BM + 1];  // +1: conflict-free transposed staging
-  __shared__ float Bs[BK][BN];
+  // double-buffered LDS: loads for tile t+1 fly under tile t's MFMAs
+  __shared__ float As[2][BK][BM + 1];
+  __shared__ float Bs[2][BK][BN];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;     // 4 waves: 2x2 of 64x64 wave tiles
   const int wr = (wave >> 1) * 64;       // wave row offset in block tile
   const int wc = (wave & 1) * 64;
+  const int tid = threadIdx.x;
 
-  const int block_row = blockIdx.x * BM;
-  const int block_col = blockIdx.y * BN;
+  // XCD-aware bijective remap (each XCD gets a contiguous grid chunk so
+  // neighbor tiles share L2-resident A/B panels — T1, cdna guide §5.5)
+  const int nwg = gridDim.x * gridDim.y;
+  const int wg = blockIdx.y * gridDim.x + blockIdx.x;
+  const int q = nwg / 8, rr = nwg % 8;
+  const int xcd = wg % 8, idx = wg / 8;
+  const int swg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q)
+                  + idx;
+  const int block_row = (swg % gridDim.x) * BM;
+  const int block_col = (swg / gridDim.x) * BN;
 
   f32x16 acc[2][2] = {};
 
   const int ktiles = (K + BK - 1) / BK;
+  StageRegs regs;
+  stage_load(A, B, M, N, K, block_row, block_col, 0, tid, regs);
+  stage_write(As[0], Bs[0], tid, regs);
+  int cur = 0;
   for (int kt = 0; kt < ktiles; ++kt) {
-    const int k0 = kt * BK;
-    // --- stage A (transposed): thread t loads A[m][4k4..4k4+3] as float4 ---
-    {
-      const int k4 = threadIdx.x & 7;          // 8 float4 per 32-wide K row
-      const int m0 = threadIdx.x >> 3;         // 32 rows per pass
-      for (int p = 0; p < 4; ++p) {
-        const int m = m0 + p * 32;
-        const int gm = block_row + m;
-        const int gk = k0 + k4 * 4;
-        float4 v = {0.f, 0.f, 0.f, 0.f};
-        if (gm < M) {
-          if (gk + 3 < K) {
-            v = *reinterpret_cast<const float4*>(A + (long)gm * K + gk);
-          } else {
-            const float* row = A + (long)gm * K;
-            if (gk + 0 < K) v.x = row[gk + 0];
-            if (gk + 1 < K) v.y = row[gk + 1];
-            if (gk + 2 < K) v.z = row[gk + 2];
-            if (gk + 3 < K) v.w = row[gk + 3];
-          }
-        }
-        As[k4 * 4 + 0][m] = v.x;
-        As[k4 * 4 + 1][m] = v.y;
-        As[k4 * 4 + 2][m] = v.z;
-        As[k4 * 4 + 3][m] = v.w;
-      }
-    }
-    // --- stage B row-major: thread t loads B[k][4n4..] as float4 ---
-    {
-      const int n4 = threadIdx.x & 31;         // 32 float4 per 128-wide row
-      const int kk0 = threadIdx.x >> 5;        // 8 k rows per pass
-      for (int p = 0; p < 4; ++p) {
-        const int kk = kk0 + p * 8;
-        const int gk = k0 + kk;
-        const int gn = block_col + n4 * 4;
-        float4 v = {0.f, 0.f, 0.f, 0.f};
-        if (gk < K) {
-          if (gn + 3 < N) {
-            v = *reinterpret_cast<const float4*>(B + (long)gk * N + gn);
-          } else {
-            const float* row = B + (long)gk * N;
-            if (gn + 0 < N) v.x = row[gn + 0];
-            if (gn + 1 < N) v.y = row[gn + 1];
-            if (gn + 2 < N) v.z = row[gn + 2];
-            if (gn + 3 < N) v.w = row[gn + 3];
-          }
-        }
-        *reinterpret_cast<float4*>(&Bs[kk][n4 * 4]) = v;
-      }
-    }
-    __syncthreads();
+    __syncthreads();  // buf[cur] staged and visible
+    if (kt + 1 < ktiles)  // issue next tile's global loads now ...
+      stage_load(A, B, M, N, K, block_row, block_col, (kt + 1) * BK, tid,
+                 regs);
 
-    // --- MFMA inner loop: K advances 2 per mfma_f32_32x32x2_f32 ---
     const int l31 = lane & 31;
     const int khalf = lane >> 5;  // this lane's k within the 2-wide step
     for (int kk = 0; kk < BK; kk += 2) {
-      const float a0 = As[kk + khalf][wr + l31];
-      const float a1 = As[kk + khalf][wr + 32 + l31];
-      const float b0 = Bs[kk + khalf][wc + l31];
-      const float b1 = Bs[kk + khalf][wc + 32 + l31];
+      const float a0 = As[cur][kk + khalf][wr + l31];
+      const float a1 = As[cur][kk + khalf][wr + 32 + l31];
+      const float b0 = Bs[cur][kk + khalf][wc + l31];
+      const float b1 = Bs[cur][kk + khalf][wc + 32 + l31];
       acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
       acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
       acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
       acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
     }
-    __syncthreads();
+    if (kt + 1 < ktiles)  // ... and land them in the other buffer (safe:
+      stage_write(As[cur ^ 1], Bs[cur ^ 1], tid, regs);  // last read of
+    cur ^= 1;             // buf[cur^1] was before this iteration's barrier
   }
 
   // --- epilogue: C/D layout col=lane&31, row=(r&3)+8*(r>>2)+4*(lane>>5) ---
