@@ -50,6 +50,9 @@ def parse_args():
     p.add_argument("--fused-bn", dest="fused_bn",
                    action=argparse.BooleanOptionalAction, default=True,
                    help="fused BN(+add)+ReLU gfx950 kernels")
+    p.add_argument("--overlap", type=int, default=4,
+                   help="sub-buckets for backward-overlapped reduction "
+                        "(0 = synchronous single-bucket pipeline)")
     p.add_argument("--channels-last", dest="channels_last",
                    action=argparse.BooleanOptionalAction, default=False,
                    help="NHWC memory format.  Measured 20x SLOWER for fp32 "
@@ -94,7 +97,8 @@ def main():
                           momentum=0.9, weight_decay=1e-4)
     step = LPTrainStep(dm, opt, grad_exp=args.grad_exp, grad_man=args.grad_man,
                        use_APS=not args.no_aps, use_kahan=args.use_kahan,
-                       emulate_node=args.emulate_node, mode=args.mode)
+                       emulate_node=args.emulate_node, mode=args.mode,
+                       overlap=args.overlap if args.emulate_node == 1 else 0)
 
     # synthetic data: a small pool of fixed random batches resident on device
     g = torch.Generator().manual_seed(42 + rank)

@@ -22,7 +22,7 @@ class LPTrainStep:
 
     def __init__(self, model: DistModule, optimizer, *, grad_exp=4, grad_man=3,
                  use_APS=True, use_kahan=False, emulate_node=1, mode="ring",
-                 use_master=True, distributed=None):
+                 use_master=True, distributed=None, overlap=0):
         if model.bucket is None:
             assert emulate_node == 1, \
                 "emulate_node > 1 needs the fused (fp32) bucket path"
@@ -41,6 +41,12 @@ class LPTrainStep:
             distributed = dist.is_available() and dist.is_initialized() and \
                 dist.get_world_size() > 1
         self.distributed = distributed
+        self.pipeline = None
+        if overlap and model.bucket is not None and emulate_node == 1:
+            from ..parallel.overlap import OverlapPipeline
+            self.pipeline = OverlapPipeline(
+                model.bucket, grad_exp, grad_man, use_APS=use_APS,
+                use_kahan=use_kahan, mode=mode, num_buckets=overlap)
 
     def loss_scale_denom(self):
         """The reference pre-divides the loss by world*emulate so the SUM
@@ -51,17 +57,22 @@ class LPTrainStep:
     def substep(self, loss):
         """Backward + (maybe) reduce/step.  Returns True when the optimizer
         stepped (boundary micro-batch)."""
-        loss.backward()
-        if self.emulator is not None:
-            self.emulator.store_microbatch()
-            if not self.emulator.full():
-                return False
-            self.emulator.reduce_(use_APS=self.use_APS,
-                                  grad_exp=self.grad_exp,
-                                  grad_man=self.grad_man)
-        sum_gradients(self.model, use_APS=self.use_APS,
-                      grad_exp=self.grad_exp, grad_man=self.grad_man,
-                      use_kahan=self.use_kahan, mode=self.mode)
+        if self.pipeline is not None:
+            self.pipeline.begin_step()
+            loss.backward()
+            self.pipeline.finish()  # reduction overlapped with backward
+        else:
+            loss.backward()
+            if self.emulator is not None:
+                self.emulator.store_microbatch()
+                if not self.emulator.full():
+                    return False
+                self.emulator.reduce_(use_APS=self.use_APS,
+                                      grad_exp=self.grad_exp,
+                                      grad_man=self.grad_man)
+            sum_gradients(self.model, use_APS=self.use_APS,
+                          grad_exp=self.grad_exp, grad_man=self.grad_man,
+                          use_kahan=self.use_kahan, mode=self.mode)
         if self.master is not None:
             self.master.grads_from_model()
             self.optimizer.step()
