@@ -174,6 +174,7 @@ double bench(const float* dA, const float* dB, float* dC, int Nsz, int reps) {
 int main(int argc, char** argv) {
   const int Nsz = argc > 1 ? atoi(argv[1]) : 4096;
   const int reps = argc > 2 ? atoi(argv[2]) : 10;
+  const int rounds = argc > 3 ? atoi(argv[3]) : 3;
   std::vector<float> hA((long)Nsz * Nsz), hB((long)Nsz * Nsz);
   srand(1);
   for (auto& v : hA) v = (rand() / (float)RAND_MAX) * 2 - 1;
@@ -203,7 +204,7 @@ int main(int argc, char** argv) {
     printf("refcheck ok\n");
   }
 
-  for (int round = 0; round < 3; ++round) {
+  for (int round = 0; round < rounds; ++round) {
     printf("round %d: v0=%6.1f v1=%6.1f v2=%6.1f v3=%6.1f TF\n", round,
            bench<0>(dA, dB, dC, Nsz, reps), bench<1>(dA, dB, dC, Nsz, reps),
            bench<2>(dA, dB, dC, Nsz, reps), bench<3>(dA, dB, dC, Nsz, reps));
