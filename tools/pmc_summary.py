@@ -12,7 +12,7 @@ def main(paths):
     for path in paths:
         with open(path) as f:
             for row in csv.DictReader(f):
-                name = re.sub(r"[(<].*", "", row["Kernel_Name"]).strip()
+                name = re.sub(r"[(<].*", "", row.get("Kernel_Name", row.get("Kernel-Name","?"))).strip()
                 agg[name][row["Counter_Name"]].append(float(row["Counter_Value"]))
     for kernel, counters in sorted(agg.items()):
         print(kernel[:80])
