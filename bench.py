@@ -53,6 +53,8 @@ def parse_args():
     p.add_argument("--overlap", type=int, default=4,
                    help="sub-buckets for backward-overlapped reduction "
                         "(0 = synchronous single-bucket pipeline)")
+    p.add_argument("--torch-profile", default=None,
+                   help="write a torch.profiler chrome trace of 3 steps here")
     p.add_argument("--channels-last", dest="channels_last",
                    action=argparse.BooleanOptionalAction, default=False,
                    help="NHWC memory format.  Measured 20x SLOWER for fp32 "
@@ -132,6 +134,16 @@ def main():
     for i in range(args.warmup):
         one_step(i)
     sync()
+
+    if args.torch_profile and rank == 0:
+        from torch.profiler import ProfilerActivity, profile
+
+        with profile(activities=[ProfilerActivity.CPU,
+                                 ProfilerActivity.CUDA]) as prof:
+            for i in range(3):
+                one_step(1000 + i)
+            sync()
+        prof.export_chrome_trace(args.torch_profile)
     t0 = time.perf_counter()
     for i in range(args.steps):
         one_step(args.warmup + i)

@@ -50,6 +50,8 @@ def parse_args(argv=None):
     p.add_argument('-e', '--evaluate', action='store_true')
     p.add_argument('--emulate_node', default=1, type=int)
     p.add_argument('--mode', choices=['ring', 'sequential'], default='ring')
+    p.add_argument('--overlap', type=int, default=0,
+                   help='sub-buckets for backward-overlapped reduction')
     p.add_argument('--synthetic', action='store_true',
                    help='synthetic CIFAR-shaped data (no dataset on disk)')
     p.add_argument('--data-root', default='./data/cifar-10-batches-py')
@@ -112,7 +114,8 @@ def main(argv=None):
     step = LPTrainStep(dm, optimizer, grad_exp=args.grad_exp,
                        grad_man=args.grad_man, use_APS=args.use_APS,
                        use_kahan=args.use_kahan,
-                       emulate_node=args.emulate_node, mode=args.mode)
+                       emulate_node=args.emulate_node, mode=args.mode,
+                       overlap=args.overlap if args.emulate_node == 1 else 0)
 
     if args.synthetic or not os.path.isdir(args.data_root):
         if not args.synthetic and rank == 0:
