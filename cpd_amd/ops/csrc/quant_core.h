@@ -69,49 +69,56 @@ CPD_HD uint32_t round_mantissa_rne(uint32_t man, int man_bits) {
   return (man & ~(unit - 1)) + (up ? unit : 0);
 }
 
+CPD_HD int clamp_i(int v, int lo, int hi) {
+  v = v < lo ? lo : v;  // lowers to v_med3_i32 on gfx950
+  return v > hi ? hi : v;
+}
+
 // Exact power-of-two scaling m * 2^e2 for integer m <= 2^24 and any target
 // value representable in fp32: split the exponent into <= two factors each
 // in the fp32-normal range; every intermediate is exact, the final multiply
 // is correctly rounded = exact (the target-format grid is a subset of fp32).
 CPD_HD float scale_pow2(uint32_t m, int e2) {
-  const int a = e2 < -126 ? -126 : (e2 > 127 ? 127 : e2);
+  const int a = clamp_i(e2, -126, 127);
   const int b = e2 - a;  // in [-126, 127] whenever m != 0 in-range
   float r = (float)m * bits_f32((uint32_t)(a + 127) << 23);
   r *= bits_f32((uint32_t)(b + 127) << 23);  // b == 0 -> exact *1.0
   return r;
 }
 
-// Branchless except wave-uniform conditions (man_bits/exp_bits are kernel
-// arguments): data-dependent cases resolve via selects, so a wave never
-// serializes both sides of the normal/subnormal/special paths.  Semantics
-// documented above; bit-equality with the numpy oracle is tested over ~1M
-// random bit patterns per format on CPU and GPU.
+// Branchless and select-light (the cast is the inner loop of the quantized
+// GEMM, so VALU count is the cost model):
+//   * the normal/subnormal split is one clamped shift (clamp(1-new_e,0,63)
+//     is 0 on the normal path, and >= 24 naturally flushes the mantissa);
+//   * round-to-nearest-even is the arithmetic identity
+//     (man + half - 1 + lsb) & ~(unit-1) — no compares;
+//   * +-0/Inf/NaN passthrough is the single unsigned test
+//     (|x|bits - 1) >= 0x7F7FFFFF.
+// man_bits/exp_bits are wave-uniform kernel arguments, so their derived
+// constants live in SGPRs.  Bit-equality with the numpy oracle is tested
+// over ~10M random bit patterns per format on CPU and GPU.
 CPD_HD float cast_fp(float x, int man_bits, int exp_bits) {
   const uint32_t u = f32_bits(x);
-  const uint32_t exp_f = (u >> 23) & 0xFFu;
-  const uint32_t man_f = u & 0x7FFFFFu;
+  const uint32_t au = u & 0x7FFFFFFFu;
   const uint32_t sign = u & 0x80000000u;
-
-  const int true_exp = (int)exp_f - 127;
+  const int exp_f = (int)(au >> 23);
   const int bias = (1 << (exp_bits - 1)) - 1;
-  const int new_e = true_exp + bias;
-  const bool passthru = (exp_f == 0xFFu) || (u << 1) == 0u;  // Inf/NaN/+-0
-  const bool flush = (exp_f == 0u);        // fp32 subnormal -> +0
-  const bool ovf = new_e >= (1 << exp_bits) - 1;  // pre-round saturate
-  const bool sub = new_e <= 0;             // target-subnormal
+  const int new_e = exp_f - 127 + bias;
 
-  // subnormal pre-shift (sticky discarded); shift==0 on the normal path
-  int shift = sub ? 1 - new_e : 0;
-  uint32_t man = man_f | (1u << 23);
-  man = (shift > 31) ? 0u : (man >> (shift & 31));
-  man = round_mantissa_rne(man, man_bits);
-  const int out_e = sub ? 1 - bias : true_exp;
-
-  float mag = scale_pow2(man, out_e - 23);
-  mag = ovf ? bits_f32(0x7F800000u) : mag;
+  // target-subnormal pre-shift (sticky discarded); 0 on the normal path
+  const int shift = clamp_i(1 - new_e, 0, 63);
+  uint64_t man = (uint64_t)((au & 0x7FFFFFu) | 0x800000u) >> shift;
+  if (man_bits < 23) {  // uniform condition: scalar branch
+    const int drop = 23 - man_bits;
+    const uint64_t unit = 1ull << drop;
+    man = (man + (unit >> 1) - 1 + ((man >> drop) & 1)) & ~(unit - 1);
+  }
+  const int out_e = new_e > 0 ? exp_f - 127 : 1 - bias;
+  float mag = scale_pow2((uint32_t)man, out_e - 23);
+  mag = new_e >= (1 << exp_bits) - 1 ? bits_f32(0x7F800000u) : mag;  // ovf
   float res = bits_f32(f32_bits(mag) | sign);
-  res = flush ? 0.0f : res;
-  return passthru ? x : res;
+  res = exp_f == 0 ? 0.0f : res;              // fp32 subnormal flush
+  return (au - 1u >= 0x7F7FFFFFu) ? x : res;  // +-0 / Inf / NaN passthrough
 }
 
 // One step of (exp,man)-rounded Kahan compensated accumulation:
