@@ -75,6 +75,51 @@ Tensor kahan_qadd_(Tensor acc, Tensor comp, const Tensor& inc, int64_t man_bits,
   return acc;
 }
 
+// bf16-wire ring hops (CPU mirror of the GPU kernels; exact for values on an
+// (exp<=8, man<=7) grid — low 16 mantissa bits are zero).
+inline float bf16_to_f32(uint16_t h) {
+  return cpd::bits_f32((uint32_t)h << 16);
+}
+inline uint16_t f32_to_bf16_exact(float f) {
+  return (uint16_t)(cpd::f32_bits(f) >> 16);
+}
+
+Tensor qadd_bf16_(Tensor acc, const Tensor& inc, int64_t man_bits,
+                  int64_t exp_bits) {
+  TORCH_CHECK(acc.scalar_type() == at::kBFloat16 && acc.is_contiguous());
+  TORCH_CHECK(inc.scalar_type() == at::kBFloat16 && inc.is_contiguous());
+  TORCH_CHECK(man_bits <= 7, "bf16 wire is exact only for man_bits <= 7");
+  TORCH_CHECK(acc.numel() == inc.numel(), "size mismatch");
+  auto* a = reinterpret_cast<uint16_t*>(acc.data_ptr<at::BFloat16>());
+  const auto* g = reinterpret_cast<const uint16_t*>(inc.data_ptr<at::BFloat16>());
+  at::parallel_for(0, acc.numel(), kGrain, [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e; ++i)
+      a[i] = f32_to_bf16_exact(cpd::cast_fp(
+          bf16_to_f32(a[i]) + bf16_to_f32(g[i]), (int)man_bits, (int)exp_bits));
+  });
+  return acc;
+}
+
+Tensor kahan_qadd_bf16_(Tensor acc, Tensor comp, const Tensor& inc,
+                        int64_t man_bits, int64_t exp_bits) {
+  TORCH_CHECK(acc.scalar_type() == at::kBFloat16 && acc.is_contiguous());
+  TORCH_CHECK(man_bits <= 7, "bf16 wire is exact only for man_bits <= 7");
+  TORCH_CHECK(acc.numel() == inc.numel() && comp.numel() == acc.numel());
+  auto* a = reinterpret_cast<uint16_t*>(acc.data_ptr<at::BFloat16>());
+  auto* c = reinterpret_cast<uint16_t*>(comp.data_ptr<at::BFloat16>());
+  const auto* g = reinterpret_cast<const uint16_t*>(inc.data_ptr<at::BFloat16>());
+  at::parallel_for(0, acc.numel(), kGrain, [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e; ++i) {
+      float af = bf16_to_f32(a[i]), cf = bf16_to_f32(c[i]);
+      cpd::kahan_qstep(af, cf, bf16_to_f32(g[i]), (int)man_bits,
+                       (int)exp_bits);
+      a[i] = f32_to_bf16_exact(af);
+      c[i] = f32_to_bf16_exact(cf);
+    }
+  });
+  return acc;
+}
+
 // Per-segment APS max-exponent scan over a flat gradient buffer:
 //   out[s] = ceil(log2(max_i |x_i| * world_size)) for segment s,
 //   -100 when the segment is all-zero (mix.py:260 sentinel).
@@ -186,6 +231,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_", &quantize_, "FP32 -> (exp,man) grid, in-place");
   m.def("qadd_", &qadd_, "acc = Q(acc + inc)");
   m.def("kahan_qadd_", &kahan_qadd_, "quantized Kahan accumulate step");
+  m.def("qadd_bf16_", &qadd_bf16_, "bf16-wire quantized accumulate");
+  m.def("kahan_qadd_bf16_", &kahan_qadd_bf16_, "bf16-wire Kahan step");
   m.def("seg_max_exp", &seg_max_exp, "per-segment APS max exponent");
   m.def("scale_quantize_", &scale_quantize_, "fused per-segment scale+cast");
   m.def("seg_scale_", &seg_scale_, "per-segment power-of-two scale");

@@ -181,3 +181,32 @@ def test_seg_aligned_matches_generic():
     ops.seg_scale_(want, offsets_c, shifts_c, -1)
     ops.seg_scale_(got_a, offsets_g, shifts_c.cuda(), -1, aligned=True)
     assert torch.equal(got_a.cpu(), want)
+
+
+def test_kernel_determinism():
+    """Race guard: every reduction kernel must be bit-deterministic across
+    repeated runs on identical inputs (no float atomics, fixed-shape
+    two-level reductions by design)."""
+    torch.manual_seed(11)
+    n = 1 << 22
+    x = (torch.randn(n, device="cuda") * 3).contiguous()
+    offsets = torch.arange(0, n + 1, n // 32, dtype=torch.int64,
+                           device="cuda")
+    r1 = ops.seg_max_exp(x, offsets, 8, aligned=True)
+    r2 = ops.seg_max_exp(x, offsets, 8, aligned=True)
+    assert torch.equal(r1, r2)
+
+    from cpd_amd.models.fused_bn import FusedBNReLU
+    m = FusedBNReLU(64).cuda().train()
+    xb = torch.randn(32, 64, 16, 16, device="cuda")
+    outs = []
+    for _ in range(2):
+        m2 = FusedBNReLU(64).cuda().train()
+        m2.load_state_dict(m.state_dict())
+        y = m2(xb)
+        y.sum().backward()
+        outs.append((y.detach().clone(), m2.weight.grad.clone(),
+                     m2.bias.grad.clone()))
+    assert torch.equal(outs[0][0], outs[1][0])
+    assert torch.equal(outs[0][1], outs[1][1])
+    assert torch.equal(outs[0][2], outs[1][2])
