@@ -83,7 +83,10 @@ def main():
     from cpd_amd.trainers.core import LPTrainStep
 
     torch.manual_seed(1234)
-    torch.backends.cudnn.benchmark = True  # MIOpen find: tune conv algos once
+    # MIOpen find: tune conv algos once during warmup (disable with
+    # CPD_BENCHMARK_FIND=0, e.g. for clean kernel-trace profiles)
+    torch.backends.cudnn.benchmark = \
+        os.environ.get("CPD_BENCHMARK_FIND", "1") == "1"
     shapes = {
         "resnet18_cifar": ((3, 32, 32), 10),
         "resnet50": ((3, 224, 224), 1000),

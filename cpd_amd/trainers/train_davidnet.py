@@ -151,6 +151,12 @@ def run_epoch(args, loader, model, step, optimizer, lr_sched, epoch, device,
             steps += 1
             if args.steps_per_epoch and steps >= args.steps_per_epoch:
                 break
+    # cross-rank stat reduction (reference `collect`, utils.py:347-356)
+    import torch.distributed as dist
+    if dist.is_available() and dist.is_initialized():
+        t = torch.tensor([tot_loss, tot_correct, float(n)])
+        dist.all_reduce(t)
+        tot_loss, tot_correct, n = float(t[0]), float(t[1]), int(t[2])
     return {'loss': tot_loss / max(n, 1), 'acc': tot_correct / max(n, 1)}
 
 
