@@ -50,9 +50,11 @@ def parse_args():
     p.add_argument("--fused-bn", dest="fused_bn",
                    action=argparse.BooleanOptionalAction, default=True,
                    help="fused BN(+add)+ReLU gfx950 kernels")
-    p.add_argument("--overlap", type=int, default=4,
+    p.add_argument("--overlap", type=int, default=0,
                    help="sub-buckets for backward-overlapped reduction "
-                        "(0 = synchronous single-bucket pipeline)")
+                        "(0 = synchronous single-bucket pipeline; overlap is "
+                        "validated multi-process on gloo but not yet on an "
+                        "8-GPU RCCL node, so the unattended default is sync)")
     p.add_argument("--torch-profile", default=None,
                    help="write a torch.profiler chrome trace of 3 steps here")
     p.add_argument("--channels-last", dest="channels_last",

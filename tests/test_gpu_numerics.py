@@ -210,3 +210,32 @@ def test_kernel_determinism():
     assert torch.equal(outs[0][0], outs[1][0])
     assert torch.equal(outs[0][1], outs[1][1])
     assert torch.equal(outs[0][2], outs[1][2])
+
+
+def test_overlap_pipeline_single_gpu():
+    """Exercise the backward-overlapped reducer's GPU path (streams, events,
+    hook-launched kernels) at W=1 and check equality with the sync path."""
+    from cpd_amd.parallel import DistModule
+    from cpd_amd.trainers.core import LPTrainStep
+
+    crit = torch.nn.CrossEntropyLoss()
+    outs = {}
+    for tag, overlap in (("sync", 0), ("overlap", 3)):
+        torch.manual_seed(4)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(64, 128), torch.nn.ReLU(),
+            torch.nn.Linear(128, 8)).cuda()
+        dm = DistModule(model)
+        opt = torch.optim.SGD([{"params": model.parameters()}], lr=0.1)
+        step = LPTrainStep(dm, opt, grad_exp=4, grad_man=3, use_APS=True,
+                           overlap=overlap)
+        gen = torch.Generator().manual_seed(2)
+        for _ in range(3):
+            x = torch.randn(16, 64, generator=gen).cuda()
+            y = torch.randint(0, 8, (16,), generator=gen).cuda()
+            step.substep(crit(dm(x), y))
+        torch.cuda.synchronize()
+        outs[tag] = {n: p.detach().cpu().clone()
+                     for n, p in model.named_parameters()}
+    for name in outs["sync"]:
+        assert torch.equal(outs["sync"][name], outs["overlap"][name]), name
