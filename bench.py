@@ -57,6 +57,10 @@ def parse_args():
                         "8-GPU RCCL node, so the unattended default is sync)")
     p.add_argument("--torch-profile", default=None,
                    help="write a torch.profiler chrome trace of 3 steps here")
+    p.add_argument("--hip-graph", dest="hip_graph",
+                   action=argparse.BooleanOptionalAction, default=False,
+                   help="capture the whole training step in a hipGraph "
+                        "(single-GPU, emulate_node=1 only)")
     p.add_argument("--channels-last", dest="channels_last",
                    action=argparse.BooleanOptionalAction, default=False,
                    help="NHWC memory format.  Measured 20x SLOWER for fp32 "
@@ -122,11 +126,35 @@ def main():
     criterion = torch.nn.CrossEntropyLoss().to(device)
     denom = step.loss_scale_denom()
 
-    def one_step(i):
-        for mb in range(args.emulate_node):
-            x, y = pool[(i * args.emulate_node + mb) % len(pool)]
-            loss = criterion(model(x), y) / denom
-            step.substep(loss)
+    use_graph = (args.hip_graph and use_gpu and world == 1
+                 and args.emulate_node == 1)
+    if use_graph:
+        # whole-step capture: fwd + bwd + APS/quantize pipeline + master
+        # update + SGD, replayed with only an input copy per step (the
+        # pipeline is host-sync-free by construction, so it captures clean)
+        static_x = pool[0][0].clone()
+        static_y = pool[0][1].clone()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                step.substep(criterion(model(static_x), static_y) / denom)
+        torch.cuda.current_stream().wait_stream(side)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            step.substep(criterion(model(static_x), static_y) / denom)
+
+        def one_step(i):
+            x, y = pool[i % len(pool)]
+            static_x.copy_(x)
+            static_y.copy_(y)
+            graph.replay()
+    else:
+        def one_step(i):
+            for mb in range(args.emulate_node):
+                x, y = pool[(i * args.emulate_node + mb) % len(pool)]
+                loss = criterion(model(x), y) / denom
+                step.substep(loss)
 
     def sync():
         if use_gpu:
