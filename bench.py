@@ -109,10 +109,13 @@ def main():
     # overlap pays only when there is communication to hide (measured -2.5%
     # at N=1 from stream/event overhead, wins at N>1)
     overlap = args.overlap if (args.emulate_node == 1 and world > 1) else 0
+    # fp32 model: stepping the params directly is bitwise-identical to the
+    # master-copy path (masters exist for low-precision models) and saves two
+    # full passes over the 45 MB bucket per step
     step = LPTrainStep(dm, opt, grad_exp=args.grad_exp, grad_man=args.grad_man,
                        use_APS=not args.no_aps, use_kahan=args.use_kahan,
                        emulate_node=args.emulate_node, mode=args.mode,
-                       overlap=overlap)
+                       overlap=overlap, use_master=False)
 
     # synthetic data: a small pool of fixed random batches resident on device
     g = torch.Generator().manual_seed(42 + rank)

@@ -20,12 +20,15 @@ class _FusedBNFn(torch.autograd.Function):
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum,
                 eps, relu, residual):
         ext = ops.hip_ext()
-        y, mean, invstd = ext.bn_relu_fwd(
+        y, mean, invstd, mask = ext.bn_relu_fwd(
             x, weight, bias, running_mean, running_var, momentum, eps, relu,
             residual)
         ctx.relu = relu
         ctx.has_res = residual is not None
-        if relu:
+        ctx.has_mask = mask is not None and mask.numel() > 0
+        if ctx.has_mask:        # 1-bit/elem ReLU mask: bwd skips the y read
+            ctx.save_for_backward(x, weight, mean, invstd, mask)
+        elif relu:
             ctx.save_for_backward(x, weight, mean, invstd, y)
         else:
             ctx.save_for_backward(x, weight, mean, invstd)
@@ -33,14 +36,16 @@ class _FusedBNFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        if ctx.relu:
+        mask = y = None
+        if ctx.has_mask:
+            x, weight, mean, invstd, mask = ctx.saved_tensors
+        elif ctx.relu:
             x, weight, mean, invstd, y = ctx.saved_tensors
         else:
             x, weight, mean, invstd = ctx.saved_tensors
-            y = None
         ext = ops.hip_ext()
         dx, dgamma, dbeta, dres = ext.bn_relu_bwd(
-            x, dy, y, mean, invstd, weight, ctx.has_res)
+            x, dy, y, mean, invstd, weight, ctx.has_res, mask)
         return (dx, dgamma, dbeta, None, None, None, None, None,
                 dres if ctx.has_res else None)
 

@@ -124,6 +124,150 @@ __global__ void bn_norm_kernel(const float* __restrict__ x,
   }
 }
 
+// ---- mask-variant forward pass 2 (8 elems/thread, HW % 8 == 0): also emits
+// the ReLU mask as 1 bit/element so backward never re-reads y (saves ~30%
+// of the backward HBM traffic) ----------------------------------------------
+__global__ void bn_norm_mask_kernel(const float* __restrict__ x,
+                                    const float* __restrict__ res,
+                                    float* __restrict__ y,
+                                    unsigned char* __restrict__ mask,
+                                    int N, int C, long HW,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ gamma,
+                                    const float* __restrict__ beta) {
+  const long total8 = (long)N * C * HW / 8;
+  const long stride = (long)gridDim.x * TPB;
+  for (long i8 = (long)blockIdx.x * TPB + threadIdx.x; i8 < total8;
+       i8 += stride) {
+    const long i = i8 * 8;
+    const int c = (int)((i / HW) % C);
+    const float a = gamma[c] * invstd[c];
+    const float b = beta[c] - mean[c] * a;
+    float4 v0 = ld4(x + i);
+    float4 v1 = ld4(x + i + 4);
+    v0.x = v0.x * a + b; v0.y = v0.y * a + b;
+    v0.z = v0.z * a + b; v0.w = v0.w * a + b;
+    v1.x = v1.x * a + b; v1.y = v1.y * a + b;
+    v1.z = v1.z * a + b; v1.w = v1.w * a + b;
+    if (res) {
+      const float4 r0 = ld4(res + i);
+      const float4 r1 = ld4(res + i + 4);
+      v0.x += r0.x; v0.y += r0.y; v0.z += r0.z; v0.w += r0.w;
+      v1.x += r1.x; v1.y += r1.y; v1.z += r1.z; v1.w += r1.w;
+    }
+    unsigned m = (v0.x > 0.f) | ((v0.y > 0.f) << 1) | ((v0.z > 0.f) << 2) |
+                 ((v0.w > 0.f) << 3) | ((v1.x > 0.f) << 4) |
+                 ((v1.y > 0.f) << 5) | ((v1.z > 0.f) << 6) |
+                 ((v1.w > 0.f) << 7);
+    v0.x = fmaxf(v0.x, 0.f); v0.y = fmaxf(v0.y, 0.f);
+    v0.z = fmaxf(v0.z, 0.f); v0.w = fmaxf(v0.w, 0.f);
+    v1.x = fmaxf(v1.x, 0.f); v1.y = fmaxf(v1.y, 0.f);
+    v1.z = fmaxf(v1.z, 0.f); v1.w = fmaxf(v1.w, 0.f);
+    st4(y + i, v0);
+    st4(y + i + 4, v1);
+    mask[i8] = (unsigned char)m;
+  }
+}
+
+// ---- mask-variant backward reduce (8 elems/thread) ------------------------
+__global__ void bn_bwd_reduce_mask_kernel(const float* __restrict__ x,
+                                          const float* __restrict__ dy,
+                                          const unsigned char* __restrict__ mask,
+                                          int N, int C, long HW, int split,
+                                          const float* __restrict__ mean,
+                                          const float* __restrict__ invstd,
+                                          float* __restrict__ ws) {
+  const int c = blockIdx.x;
+  const int s = blockIdx.y;
+  __shared__ float smem[TPB / 64];
+  const float m = mean[c], is = invstd[c];
+  float sd = 0.f, sdx = 0.f;
+  for (int n = s; n < N; n += split) {
+    const long off = ((long)n * C + c) * HW;
+    for (long i = threadIdx.x * 8; i < HW; i += (long)TPB * 8) {
+      float4 g0 = ld4(dy + off + i);
+      float4 g1 = ld4(dy + off + i + 4);
+      const float4 v0 = ld4(x + off + i);
+      const float4 v1 = ld4(x + off + i + 4);
+      const unsigned mk = mask[(off + i) / 8];
+      g0.x = (mk & 1) ? g0.x : 0.f;
+      g0.y = (mk & 2) ? g0.y : 0.f;
+      g0.z = (mk & 4) ? g0.z : 0.f;
+      g0.w = (mk & 8) ? g0.w : 0.f;
+      g1.x = (mk & 16) ? g1.x : 0.f;
+      g1.y = (mk & 32) ? g1.y : 0.f;
+      g1.z = (mk & 64) ? g1.z : 0.f;
+      g1.w = (mk & 128) ? g1.w : 0.f;
+      sd += g0.x + g0.y + g0.z + g0.w + g1.x + g1.y + g1.z + g1.w;
+      sdx += g0.x * (v0.x - m) + g0.y * (v0.y - m) + g0.z * (v0.z - m) +
+             g0.w * (v0.w - m) + g1.x * (v1.x - m) + g1.y * (v1.y - m) +
+             g1.z * (v1.z - m) + g1.w * (v1.w - m);
+    }
+  }
+  const float bs = block_reduce(sd, smem);
+  __syncthreads();
+  const float bq = block_reduce(sdx, smem);
+  if (threadIdx.x == 0) {
+    ws[((long)c * split + s) * 2 + 0] = bs;
+    ws[((long)c * split + s) * 2 + 1] = bq * is;
+  }
+}
+
+// ---- mask-variant backward dx (8 elems/thread) ----------------------------
+__global__ void bn_bwd_dx_mask_kernel(const float* __restrict__ x,
+                                      const float* __restrict__ dy,
+                                      const unsigned char* __restrict__ mask,
+                                      float* __restrict__ dx,
+                                      float* __restrict__ dres,
+                                      int N, int C, long HW,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ invstd,
+                                      const float* __restrict__ gamma,
+                                      const float* __restrict__ sum_dy,
+                                      const float* __restrict__ sum_dyx,
+                                      float inv_count) {
+  const long total8 = (long)N * C * HW / 8;
+  const long stride = (long)gridDim.x * TPB;
+  for (long i8 = (long)blockIdx.x * TPB + threadIdx.x; i8 < total8;
+       i8 += stride) {
+    const long i = i8 * 8;
+    const int c = (int)((i / HW) % C);
+    const float m = mean[c], is = invstd[c];
+    const float k = gamma[c] * is;
+    const float md = sum_dy[c] * inv_count;
+    const float mdx = sum_dyx[c] * inv_count;
+    float4 g0 = ld4(dy + i);
+    float4 g1 = ld4(dy + i + 4);
+    const float4 v0 = ld4(x + i);
+    const float4 v1 = ld4(x + i + 4);
+    const unsigned mk = mask[i8];
+    g0.x = (mk & 1) ? g0.x : 0.f;
+    g0.y = (mk & 2) ? g0.y : 0.f;
+    g0.z = (mk & 4) ? g0.z : 0.f;
+    g0.w = (mk & 8) ? g0.w : 0.f;
+    g1.x = (mk & 16) ? g1.x : 0.f;
+    g1.y = (mk & 32) ? g1.y : 0.f;
+    g1.z = (mk & 64) ? g1.z : 0.f;
+    g1.w = (mk & 128) ? g1.w : 0.f;
+    if (dres) {
+      st4(dres + i, g0);
+      st4(dres + i + 4, g1);
+    }
+    float4 o0, o1;
+    o0.x = k * (g0.x - md - (v0.x - m) * is * mdx);
+    o0.y = k * (g0.y - md - (v0.y - m) * is * mdx);
+    o0.z = k * (g0.z - md - (v0.z - m) * is * mdx);
+    o0.w = k * (g0.w - md - (v0.w - m) * is * mdx);
+    o1.x = k * (g1.x - md - (v1.x - m) * is * mdx);
+    o1.y = k * (g1.y - md - (v1.y - m) * is * mdx);
+    o1.z = k * (g1.z - md - (v1.z - m) * is * mdx);
+    o1.w = k * (g1.w - md - (v1.w - m) * is * mdx);
+    st4(dx + i, o0);
+    st4(dx + i + 4, o1);
+  }
+}
+
 // ---- backward pass 1: per (c, slice) partials of sum(dy_eff),
 //      sum(dy_eff * xhat); dy_eff = dy * (y > 0) when relu ----------------
 __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
@@ -277,18 +421,31 @@ std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
     TORCH_CHECK(residual->is_contiguous() && residual->sizes() == x.sizes());
     res_ptr = residual->data_ptr<float>();
   }
-  hipLaunchKernelGGL(bn_norm_kernel, dim3(elem_grid((long)N * C * HW / 4)),
-                     dim3(TPB), 0, st, x.data_ptr<float>(), res_ptr,
-                     y.data_ptr<float>(), N, C, HW, mean.data_ptr<float>(),
-                     invstd.data_ptr<float>(), gamma.data_ptr<float>(),
-                     beta.data_ptr<float>(), relu ? 1 : 0);
-  return {y, mean, invstd};
+  Tensor mask = at::empty({0}, opts.dtype(at::kByte));
+  if (relu && HW % 8 == 0) {
+    // emit the ReLU mask as 1 bit/element so backward skips the y re-read
+    mask = at::empty({(long)N * C * HW / 8}, opts.dtype(at::kByte));
+    hipLaunchKernelGGL(bn_norm_mask_kernel,
+                       dim3(elem_grid((long)N * C * HW / 8)), dim3(TPB), 0, st,
+                       x.data_ptr<float>(), res_ptr, y.data_ptr<float>(),
+                       mask.data_ptr<uint8_t>(), N, C, HW,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(bn_norm_kernel, dim3(elem_grid((long)N * C * HW / 4)),
+                       dim3(TPB), 0, st, x.data_ptr<float>(), res_ptr,
+                       y.data_ptr<float>(), N, C, HW, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), relu ? 1 : 0);
+  }
+  return {y, mean, invstd, mask};
 }
 
 std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
                                 const c10::optional<Tensor>& y_for_mask,
                                 const Tensor& mean, const Tensor& invstd,
-                                const Tensor& gamma, bool need_dres) {
+                                const Tensor& gamma, bool need_dres,
+                                const c10::optional<Tensor>& bitmask) {
   const int N = x.size(0), C = x.size(1);
   const long HW = (long)x.size(2) * x.size(3);
   const int split = pick_split(N, C, HW);
@@ -304,6 +461,33 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
   const float* yp = y_for_mask.has_value() ? y_for_mask->data_ptr<float>()
                                            : nullptr;
   Tensor dyc = dy.contiguous();
+  float* dres_ptr = nullptr;
+  if (need_dres) {
+    dres = at::empty_like(x);
+    dres_ptr = dres.data_ptr<float>();
+  }
+  if (bitmask.has_value() && bitmask->defined() &&
+      bitmask->numel() > 0) {
+    const uint8_t* mk = bitmask->data_ptr<uint8_t>();
+    hipLaunchKernelGGL(bn_bwd_reduce_mask_kernel, dim3(C, split), dim3(TPB),
+                       0, st, x.data_ptr<float>(), dyc.data_ptr<float>(), mk,
+                       N, C, HW, split, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), ws.data_ptr<float>());
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + TPB - 1) / TPB),
+                       dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
+                       sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+                       dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
+    hipLaunchKernelGGL(bn_bwd_dx_mask_kernel,
+                       dim3(elem_grid((long)N * C * HW / 8)), dim3(TPB), 0,
+                       st, x.data_ptr<float>(), dyc.data_ptr<float>(), mk,
+                       dx.data_ptr<float>(), dres_ptr, N, C, HW,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                       sum_dyx.data_ptr<float>(),
+                       1.0f / (float)((long)N * HW));
+    if (!need_dres) dres = at::Tensor();
+    return {dx, dgamma, dbeta, dres};
+  }
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C, split), dim3(TPB), 0, st,
                      x.data_ptr<float>(), dyc.data_ptr<float>(), yp, N, C, HW,
                      split, mean.data_ptr<float>(), invstd.data_ptr<float>(),
@@ -312,11 +496,6 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
                      dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
                      sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
                      dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
-  float* dres_ptr = nullptr;
-  if (need_dres) {
-    dres = at::empty_like(x);
-    dres_ptr = dres.data_ptr<float>();
-  }
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(elem_grid((long)N * C * HW / 4)),
                      dim3(TPB), 0, st, x.data_ptr<float>(),
                      dyc.data_ptr<float>(), yp, dx.data_ptr<float>(), dres_ptr,
