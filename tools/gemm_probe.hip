@@ -155,6 +155,100 @@ __global__ __launch_bounds__(256) void gemm_v(const float* __restrict__ A,
     }
 }
 
+// 256x128 tile, 8 waves (each a 2x2 of 32x32 accs like the 4-wave kernel):
+// halves B re-reads and barrier count per MFMA at the same waves/SIMD.
+constexpr int BM2 = 256, BN2 = 128;
+
+__global__ __launch_bounds__(512) void gemm_big(const float* __restrict__ A,
+                                                const float* __restrict__ B,
+                                                float* __restrict__ C, int M,
+                                                int N, int K) {
+  __shared__ float As[2][BK][BM2 + 1];
+  __shared__ float Bs[2][BK][BN2];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;         // 8 waves: 4 rows x 2 cols
+  const int wr = (wave >> 1) * 64, wc = (wave & 1) * 64;
+  const int tid = threadIdx.x;
+  const int nwg = gridDim.x * gridDim.y;
+  const int wg = blockIdx.y * gridDim.x + blockIdx.x;
+  const int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+  const int swg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  const int block_row = (swg % gridDim.x) * BM2;
+  const int block_col = (swg / gridDim.x) * BN2;
+
+  f32x16 acc[2][2] = {};
+  const int ktiles = K / BK;
+  float4 ra[4], rb[2];
+  auto load2 = [&](int k0) {
+    const int k4 = tid & 7, m0 = tid >> 3;        // A: 64 rows/pass x 4
+    for (int p = 0; p < 4; ++p)
+      ra[p] = *reinterpret_cast<const float4*>(
+          A + (long)(block_row + m0 + p * 64) * K + k0 + k4 * 4);
+    const int n4 = tid & 31, kk0 = tid >> 5;      // B: 16 k-rows/pass x 2
+    for (int p = 0; p < 2; ++p)
+      rb[p] = *reinterpret_cast<const float4*>(
+          B + (long)(k0 + kk0 + p * 16) * N + block_col + n4 * 4);
+  };
+  auto write2 = [&](int buf) {
+    const int k4 = tid & 7, m0 = tid >> 3;
+    for (int p = 0; p < 4; ++p) {
+      const int m = m0 + p * 64;
+      As[buf][k4 * 4 + 0][m] = ra[p].x;
+      As[buf][k4 * 4 + 1][m] = ra[p].y;
+      As[buf][k4 * 4 + 2][m] = ra[p].z;
+      As[buf][k4 * 4 + 3][m] = ra[p].w;
+    }
+    const int n4 = tid & 31, kk0 = tid >> 5;
+    for (int p = 0; p < 2; ++p)
+      *reinterpret_cast<float4*>(&Bs[buf][kk0 + p * 16][n4 * 4]) = rb[p];
+  };
+  load2(0);
+  write2(0);
+  int cur = 0;
+  const int l31 = lane & 31, kh = lane >> 5;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < ktiles) load2((kt + 1) * BK);
+    for (int kk = 0; kk < BK; kk += 2) {
+      const float a0 = As[cur][kk + kh][wr + l31];
+      const float a1 = As[cur][kk + kh][wr + 32 + l31];
+      const float b0 = Bs[cur][kk + kh][wc + l31];
+      const float b1 = Bs[cur][kk + kh][wc + 32 + l31];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    if (kt + 1 < ktiles) write2(cur ^ 1);
+    cur ^= 1;
+  }
+  for (int mi = 0; mi < 2; ++mi)
+    for (int nj = 0; nj < 2; ++nj) {
+      const int col = block_col + wc + nj * 32 + (lane & 31);
+      for (int r = 0; r < 16; ++r) {
+        const int row = block_row + wr + mi * 32 + (r & 3) + 8 * (r >> 2) +
+                        4 * (lane >> 5);
+        C[(long)row * N + col] = acc[mi][nj][r];
+      }
+    }
+}
+
+double bench_big(const float* dA, const float* dB, float* dC, int Nsz,
+                 int reps) {
+  dim3 grid(Nsz / BM2, Nsz / BN2), block(512);
+  hipLaunchKernelGGL(gemm_big, grid, block, 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+  HIP_CHECK(hipDeviceSynchronize());
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < reps; ++i)
+    hipLaunchKernelGGL(gemm_big, grid, block, 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+  hipEventRecord(t1);
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return 2.0 * Nsz * Nsz * (double)Nsz * reps / (ms * 1e-3) / 1e12;
+}
+
 template <int V>
 double bench(const float* dA, const float* dB, float* dC, int Nsz, int reps) {
   dim3 grid(Nsz / BM, Nsz / BN), block(256);
@@ -204,10 +298,21 @@ int main(int argc, char** argv) {
     printf("refcheck ok\n");
   }
 
+  if (Nsz % BM2 == 0) {  // bit-check big-tile vs baseline (exact fmaf chain)
+    std::vector<float> h0((long)Nsz * Nsz), h1((long)Nsz * Nsz);
+    dim3 g0(Nsz / BM, Nsz / BN), g1(Nsz / BM2, Nsz / BN2);
+    hipLaunchKernelGGL((gemm_v<0>), g0, dim3(256), 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+    HIP_CHECK(hipMemcpy(h0.data(), dC, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
+    hipLaunchKernelGGL(gemm_big, g1, dim3(512), 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+    HIP_CHECK(hipMemcpy(h1.data(), dC, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
+    long bad = 0;
+    for (long i = 0; i < (long)Nsz * Nsz; ++i) bad += (h0[i] != h1[i]);
+    printf("bigcheck: %ld mismatches\n", bad);
+  }
   for (int round = 0; round < rounds; ++round) {
-    printf("round %d: v0=%6.1f v1=%6.1f v2=%6.1f v3=%6.1f TF\n", round,
+    printf("round %d: v0=%6.1f v1=%6.1f v3=%6.1f big=%6.1f TF\n", round,
            bench<0>(dA, dB, dC, Nsz, reps), bench<1>(dA, dB, dC, Nsz, reps),
-           bench<2>(dA, dB, dC, Nsz, reps), bench<3>(dA, dB, dC, Nsz, reps));
+           bench<3>(dA, dB, dC, Nsz, reps), bench_big(dA, dB, dC, Nsz, reps));
   }
   return 0;
 }
