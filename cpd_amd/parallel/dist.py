@@ -101,8 +101,7 @@ def broadcast_params(model):
     if not (dist.is_available() and dist.is_initialized()):
         return
     for p in model.state_dict().values():
-        if p.numel() > 0 and p.is_floating_point() or p.dtype in (
-                torch.int64, torch.int32):
+        if torch.is_tensor(p) and p.numel() > 0:
             dist.broadcast(p, 0)
 
 
@@ -169,7 +168,8 @@ def _sum_gradients_fused(bucket, use_APS, grad_exp, grad_man, use_kahan,
     W = _world()
     distributed = dist.is_available() and dist.is_initialized()
 
-    if grad_exp == 8 and grad_man == 23 and not use_kahan:
+    fp32_path = grad_exp == 8 and grad_man == 23 and not use_kahan
+    if fp32_path and not use_APS:
         # full-precision path: plain (bucketed) RCCL all-reduce
         if distributed:
             dist.all_reduce(flat)
@@ -189,7 +189,13 @@ def _sum_gradients_fused(bucket, use_APS, grad_exp, grad_man, use_kahan,
         if wire is None and flat.is_cuda and grad_man <= 7:
             wire = "bf16"  # values are on-grid after scale_quantize_
 
-    if distributed and W > 1:
+    if fp32_path:
+        # APS at (8,23): scaled grads sum in full precision with a plain
+        # all-reduce (reference: normal_sum_gradients' (8,23) shortcut,
+        # dist_util.py:55-59, after the APS scale+flush)
+        if distributed:
+            dist.all_reduce(flat)
+    elif distributed and W > 1:
         lp_all_reduce_(flat, grad_exp, grad_man, use_kahan=use_kahan,
                        mode=mode, wire=wire)
     else:

@@ -106,7 +106,8 @@ class OverlapPipeline:
         W = self._world()
         distributed = dist.is_available() and dist.is_initialized()
         exp, man = self.grad_exp, self.grad_man
-        if exp == 8 and man == 23 and not self.use_kahan:
+        fp32_path = exp == 8 and man == 23 and not self.use_kahan
+        if fp32_path and not self.use_APS:
             if distributed:
                 dist.all_reduce(flat)
             return
@@ -121,7 +122,10 @@ class OverlapPipeline:
             ops.scale_quantize_(flat, offs, shifts, man, exp, aligned=True)
             if wire is None and flat.is_cuda and man <= 7:
                 wire = "bf16"
-        if distributed and W > 1:
+        if fp32_path:
+            if distributed:
+                dist.all_reduce(flat)
+        elif distributed and W > 1:
             lp_all_reduce_(flat, exp, man, use_kahan=self.use_kahan,
                            mode=self.mode, wire=wire)
         else:
