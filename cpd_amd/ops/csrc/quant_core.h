@@ -57,18 +57,6 @@ CPD_HD float bits_f32(uint32_t u) {
   return f;
 }
 
-// Round-to-nearest-even of a 24-bit integer significand at `man_bits`
-// fractional bits kept below the implicit-one position (bit 23).
-CPD_HD uint32_t round_mantissa_rne(uint32_t man, int man_bits) {
-  if (man_bits >= 23) return man;
-  const uint32_t drop = 23 - man_bits;          // low bits to clear
-  const uint32_t unit = 1u << drop;             // value of the kept LSB
-  const uint32_t half = unit >> 1;
-  const uint32_t rem = man & (unit - 1);
-  const bool up = (rem > half) || (rem == half && (man & unit));
-  return (man & ~(unit - 1)) + (up ? unit : 0);
-}
-
 CPD_HD int clamp_i(int v, int lo, int hi) {
   v = v < lo ? lo : v;  // lowers to v_med3_i32 on gfx950
   return v > hi ? hi : v;

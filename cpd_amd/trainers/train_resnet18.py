@@ -34,6 +34,7 @@ from cpd_amd.trainers.core import LPTrainStep
 from cpd_amd.utils import (AverageMeter, DistributedGivenIterationSampler,
                            DistributedSampler, LARS, accuracy, load_state,
                            save_checkpoint)
+from cpd_amd.utils.scalars import ScalarLogger
 
 
 def parse_args(argv=None):
@@ -164,6 +165,7 @@ def train(args, train_loader, val_loader, dm, model, criterion, optimizer,
           max_iter, best_prec1):
     import torch.distributed as dist
 
+    scalars = ScalarLogger(args.save_path + '_scalars' if rank == 0 else None)
     batch_time = AverageMeter(args.print_freq)
     losses = AverageMeter(args.print_freq)
     curr_step = start_iter
@@ -196,11 +198,14 @@ def train(args, train_loader, val_loader, dm, model, criterion, optimizer,
                 print(f'Iter [{curr_step}/{max_iter}] lr {lr:.4f} '
                       f'loss {losses.avg:.4f} '
                       f'batch_time {batch_time.avg * 1000:.1f}ms', flush=True)
+                scalars.add_scalar('loss_train', losses.avg, curr_step)
+                scalars.add_scalar('lr', lr, curr_step)
             if curr_step % args.val_freq == 0:
                 prec1 = validate(val_loader, model, criterion, device,
                                  world_size, rank)
                 model.train()
                 if rank == 0:
+                    scalars.add_scalar('acc1', prec1, curr_step)
                     is_best = prec1 > best_prec1
                     best_prec1 = max(prec1, best_prec1)
                     os.makedirs(os.path.dirname(args.save_path) or '.',
