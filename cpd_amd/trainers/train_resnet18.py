@@ -29,7 +29,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
 
 from cpd_amd import models
 from cpd_amd.data import CIFAR10, SyntheticImages
-from cpd_amd.parallel import DistModule, dist_init, sum_gradients
+from cpd_amd.parallel import DistModule, dist_init
 from cpd_amd.trainers.core import LPTrainStep
 from cpd_amd.utils import (AverageMeter, DistributedGivenIterationSampler,
                            DistributedSampler, LARS, accuracy, load_state,

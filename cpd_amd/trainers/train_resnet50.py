@@ -5,7 +5,6 @@ sub-batch gradient accumulation (= node emulation), BatchNorm params with
 weight-decay 0, nesterov SGD with warmup/step decay, epoch checkpoints with
 auto-resume by scanning for the newest one)."""
 import argparse
-import math
 import os
 import sys
 import time

@@ -1,6 +1,5 @@
 """Fused BN(+add)+ReLU: CPU fallback equivalence (always) and GPU kernel
 numerics vs the eager composition (gpu-marked)."""
-import numpy as np
 import pytest
 import torch
 import torch.nn.functional as F

@@ -1,6 +1,5 @@
 """CPU tests for the shared training step (LPTrainStep): fp32-path equivalence
 with vanilla SGD, e4m3+APS learning progress, emulate-node cadence."""
-import numpy as np
 import torch
 import pytest
 

@@ -3,7 +3,6 @@ samplers, checkpoint round-trip with module-prefix fix-up, LARS)."""
 import os
 import tempfile
 
-import numpy as np
 import torch
 
 from cpd_amd.utils import (AverageMeter, DistributedGivenIterationSampler,

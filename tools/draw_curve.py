@@ -4,7 +4,6 @@ reference plots '* All Loss' lines with matplotlib, draw_curve.py:1-39; this
 environment has no matplotlib, so the output is TSV — plot it anywhere)."""
 import argparse
 import re
-import sys
 
 
 def parse_log(path):
