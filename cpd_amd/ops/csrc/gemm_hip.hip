@@ -203,7 +203,11 @@ __global__ __launch_bounds__(256) void gemm_f32_kernel(
 // quantized-Kahan-accumulator GEMM (VALU; rounding forbids MFMA accumulation)
 // ---------------------------------------------------------------------------
 
-constexpr int QBM = 64, QBN = 64, QBK = 16;
+// 32x32 tile, 2x2 outputs per thread: (M/32)x(N/32) blocks give 4x the
+// resident waves of a 64x64 tile — this kernel is LATENCY-bound on the
+// serial cast chains, so occupancy is the lever (64-tile: 0.34 TF, 32-tile:
+// 0.52 TF @1024^3, bit-identical — tools/quant_gemm_probe.hip).
+constexpr int QBM = 32, QBN = 32, QBK = 16;
 
 __global__ __launch_bounds__(256) void quant_gemm_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
@@ -211,30 +215,30 @@ __global__ __launch_bounds__(256) void quant_gemm_kernel(
   __shared__ float As[QBK][QBM + 1];
   __shared__ float Bs[QBK][QBN];
 
-  const int tx = threadIdx.x & 15;   // 16x16 threads, 4x4 outputs each
+  const int tx = threadIdx.x & 15;   // 16x16 threads, 2x2 outputs each
   const int ty = threadIdx.x >> 4;
-  const int row0 = blockIdx.x * QBM + ty * 4;
-  const int col0 = blockIdx.y * QBN + tx * 4;
+  const int row0 = blockIdx.x * QBM + ty * 2;
+  const int col0 = blockIdx.y * QBN + tx * 2;
 
-  float acc[4][4] = {};
-  float comp[4][4] = {};
+  float acc[2][2] = {};
+  float comp[2][2] = {};
 
   const int ktiles = (K + QBK - 1) / QBK;
   for (int kt = 0; kt < ktiles; ++kt) {
     const int k0 = kt * QBK;
-    // stage A[64][16] transposed, B[16][64]; 256 threads x 4 elements each
+    // stage A[32][16] transposed, B[16][32]; 256 threads x 2 elements each
     {
       const int k = threadIdx.x & 15;
       const int m0 = threadIdx.x >> 4;
-      for (int p = 0; p < 4; ++p) {
+      for (int p = 0; p < 2; ++p) {
         const int m = m0 + p * 16;
         const int gm = blockIdx.x * QBM + m;
         As[k][m] = (gm < M && k0 + k < K) ? A[(long)gm * K + k0 + k] : 0.0f;
       }
-      const int n = threadIdx.x & 63;
-      const int kk0 = threadIdx.x >> 6;
-      for (int p = 0; p < 4; ++p) {
-        const int kk = kk0 + p * 4;
+      const int n = threadIdx.x & 31;
+      const int kk0 = threadIdx.x >> 5;
+      for (int p = 0; p < 2; ++p) {
+        const int kk = kk0 + p * 8;
         const int gn = blockIdx.y * QBN + n;
         Bs[kk][n] = (k0 + kk < K && gn < N) ? B[(long)(k0 + kk) * N + gn] : 0.0f;
       }
@@ -243,11 +247,13 @@ __global__ __launch_bounds__(256) void quant_gemm_kernel(
 
     const int klim = min(QBK, K - k0);  // never round in padded-k steps
     for (int kk = 0; kk < klim; ++kk) {  // strictly k-ordered (semantics)
-      float a[4], b[4];
-      for (int i = 0; i < 4; ++i) a[i] = As[kk][ty * 4 + i];
-      for (int j = 0; j < 4; ++j) b[j] = Bs[kk][tx * 4 + j];
-      for (int i = 0; i < 4; ++i)
-        for (int j = 0; j < 4; ++j) {
+      float a[2], b[2];
+      a[0] = As[kk][ty * 2];
+      a[1] = As[kk][ty * 2 + 1];
+      b[0] = Bs[kk][tx * 2];
+      b[1] = Bs[kk][tx * 2 + 1];
+      for (int i = 0; i < 2; ++i)
+        for (int j = 0; j < 2; ++j) {
           const float prod = cast_fp(a[i] * b[j], man, exp);
           kahan_qstep(acc[i][j], comp[i][j], prod, man, exp);
         }
@@ -255,9 +261,9 @@ __global__ __launch_bounds__(256) void quant_gemm_kernel(
     __syncthreads();
   }
 
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < 2; ++i) {
     if (row0 + i >= M) break;
-    for (int j = 0; j < 4; ++j)
+    for (int j = 0; j < 2; ++j)
       if (col0 + j < N) C[(long)(row0 + i) * N + col0 + j] = acc[i][j];
   }
 }
