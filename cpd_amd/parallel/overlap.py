@@ -23,6 +23,12 @@ from .ring import lp_all_reduce_
 class OverlapPipeline:
     def __init__(self, bucket, grad_exp, grad_man, use_APS=True,
                  use_kahan=False, mode="ring", wire=None, num_buckets=4):
+        # one pipeline per bucket: replace (and unhook) any previous one so
+        # re-creating an LPTrainStep never double-registers hooks
+        prev = getattr(bucket, "_overlap_pipeline", None)
+        if prev is not None:
+            prev.remove()
+        bucket._overlap_pipeline = self
         self.bucket = bucket
         self.grad_exp = grad_exp
         self.grad_man = grad_man
