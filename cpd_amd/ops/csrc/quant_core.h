@@ -69,7 +69,8 @@ CPD_HD int clamp_i(int v, int lo, int hi) {
 CPD_HD float scale_pow2(uint32_t m, int e2) {
   const int a = clamp_i(e2, -126, 127);
   const int b = e2 - a;  // in [-126, 127] whenever m != 0 in-range
-  float r = (float)m * bits_f32((uint32_t)(a + 127) << 23);
+  // m <= 2^24: signed convert (vectorizes as cvtdq2ps; u32->f32 would not)
+  float r = (float)(int32_t)m * bits_f32((uint32_t)(a + 127) << 23);
   r *= bits_f32((uint32_t)(b + 127) << 23);  // b == 0 -> exact *1.0
   return r;
 }
@@ -93,16 +94,19 @@ CPD_HD float cast_fp(float x, int man_bits, int exp_bits) {
   const int bias = (1 << (exp_bits - 1)) - 1;
   const int new_e = exp_f - 127 + bias;
 
-  // target-subnormal pre-shift (sticky discarded); 0 on the normal path
-  const int shift = clamp_i(1 - new_e, 0, 63);
-  uint64_t man = (uint64_t)((au & 0x7FFFFFu) | 0x800000u) >> shift;
+  // target-subnormal pre-shift (sticky discarded); 0 on the normal path.
+  // 32-bit ops only (a u64 shift would block CPU auto-vectorization):
+  // shifts >= 32 select to zero instead of shifting.
+  const int shift = clamp_i(1 - new_e, 0, 32);
+  uint32_t man = (au & 0x7FFFFFu) | 0x800000u;
+  man = shift >= 32 ? 0u : (man >> (shift & 31));
   if (man_bits < 23) {  // uniform condition: scalar branch
     const int drop = 23 - man_bits;
-    const uint64_t unit = 1ull << drop;
+    const uint32_t unit = 1u << drop;
     man = (man + (unit >> 1) - 1 + ((man >> drop) & 1)) & ~(unit - 1);
   }
   const int out_e = new_e > 0 ? exp_f - 127 : 1 - bias;
-  float mag = scale_pow2((uint32_t)man, out_e - 23);
+  float mag = scale_pow2(man, out_e - 23);
   mag = new_e >= (1 << exp_bits) - 1 ? bits_f32(0x7F800000u) : mag;  // ovf
   float res = bits_f32(f32_bits(mag) | sign);
   res = exp_f == 0 ? 0.0f : res;              // fp32 subnormal flush
