@@ -5,6 +5,8 @@
 // GPU tests compare against (same quant_core.h compiled for both targets).
 #include <torch/extension.h>
 #include <ATen/Parallel.h>
+#include <cstring>
+#include <vector>
 
 #include "quant_core.h"
 
@@ -23,8 +25,8 @@ constexpr int64_t kGrain = 1 << 14;
 Tensor quantize(const Tensor& x, int64_t man_bits, int64_t exp_bits) {
   check_f32_contig(x, "x");
   Tensor out = at::empty_like(x);
-  const float* src = x.const_data_ptr<float>();
-  float* dst = out.mutable_data_ptr<float>();
+  const float* __restrict__ src = x.const_data_ptr<float>();
+  float* __restrict__ dst = out.mutable_data_ptr<float>();
   at::parallel_for(0, x.numel(), kGrain, [&](int64_t b, int64_t e) {
     for (int64_t i = b; i < e; ++i)
       dst[i] = cpd::cast_fp(src[i], (int)man_bits, (int)exp_bits);
@@ -48,8 +50,8 @@ Tensor qadd_(Tensor acc, const Tensor& inc, int64_t man_bits, int64_t exp_bits) 
   check_f32_contig(acc, "acc");
   check_f32_contig(inc, "inc");
   TORCH_CHECK(acc.numel() == inc.numel(), "size mismatch");
-  float* a = acc.mutable_data_ptr<float>();
-  const float* g = inc.const_data_ptr<float>();
+  float* __restrict__ a = acc.mutable_data_ptr<float>();
+  const float* __restrict__ g = inc.const_data_ptr<float>();
   at::parallel_for(0, acc.numel(), kGrain, [&](int64_t b, int64_t e) {
     for (int64_t i = b; i < e; ++i)
       a[i] = cpd::cast_fp(a[i] + g[i], (int)man_bits, (int)exp_bits);
@@ -65,9 +67,9 @@ Tensor kahan_qadd_(Tensor acc, Tensor comp, const Tensor& inc, int64_t man_bits,
   check_f32_contig(inc, "inc");
   TORCH_CHECK(acc.numel() == inc.numel() && comp.numel() == acc.numel(),
               "size mismatch");
-  float* a = acc.mutable_data_ptr<float>();
-  float* c = comp.mutable_data_ptr<float>();
-  const float* g = inc.const_data_ptr<float>();
+  float* __restrict__ a = acc.mutable_data_ptr<float>();
+  float* __restrict__ c = comp.mutable_data_ptr<float>();
+  const float* __restrict__ g = inc.const_data_ptr<float>();
   at::parallel_for(0, acc.numel(), kGrain, [&](int64_t b, int64_t e) {
     for (int64_t i = b; i < e; ++i)
       cpd::kahan_qstep(a[i], c[i], g[i], (int)man_bits, (int)exp_bits);
@@ -196,16 +198,24 @@ Tensor quant_gemm(const Tensor& a, const Tensor& b, int64_t man_bits,
   const float* B = b.const_data_ptr<float>();
   float* C = c.mutable_data_ptr<float>();
   const int mb = (int)man_bits, eb = (int)exp_bits;
+  // k must stay sequential (rounding order is the semantics); the j lanes
+  // are independent Kahan chains, so the inner loop vectorizes
   at::parallel_for(0, M, 1, [&](int64_t rb, int64_t re) {
+    std::vector<float> acc(N), comp(N);
     for (int64_t i = rb; i < re; ++i) {
-      for (int64_t j = 0; j < N; ++j) {
-        float acc = 0.0f, comp = 0.0f;
-        for (int64_t k = 0; k < K; ++k) {
-          const float prod = cpd::cast_fp(A[i * K + k] * B[k * N + j], mb, eb);
-          cpd::kahan_qstep(acc, comp, prod, mb, eb);
+      std::fill(acc.begin(), acc.end(), 0.0f);
+      std::fill(comp.begin(), comp.end(), 0.0f);
+      float* __restrict__ ac = acc.data();
+      float* __restrict__ cp = comp.data();
+      for (int64_t k = 0; k < K; ++k) {
+        const float av = A[i * K + k];
+        const float* __restrict__ Bk = B + k * N;
+        for (int64_t j = 0; j < N; ++j) {
+          const float prod = cpd::cast_fp(av * Bk[j], mb, eb);
+          cpd::kahan_qstep(ac[j], cp[j], prod, mb, eb);
         }
-        C[i * N + j] = acc;
       }
+      std::memcpy(C + i * N, ac, N * sizeof(float));
     }
   });
   return c;
