@@ -46,6 +46,46 @@ __device__ __forceinline__ void stage_write(float (*As)[BM + 1],
     *reinterpret_cast<float4*>(&Bs[kk0 + p * 8][n4 * 4]) = r.b[p];
 }
 
+__device__ __forceinline__ void stage_load_a(const float* A, int K,
+                                             int block_row, int k0, int tid,
+                                             StageRegs& r) {
+  const int k4 = tid & 7, m0 = tid >> 3;
+  for (int p = 0; p < 4; ++p)
+    r.a[p] = *reinterpret_cast<const float4*>(
+        A + (long)(block_row + m0 + p * 32) * K + k0 + k4 * 4);
+}
+
+__device__ __forceinline__ void stage_write_a(float (*As)[BM + 1], int tid,
+                                              const StageRegs& r) {
+  const int k4 = tid & 7, m0 = tid >> 3;
+  for (int p = 0; p < 4; ++p) {
+    const int m = m0 + p * 32;
+    As[k4 * 4 + 0][m] = r.a[p].x;
+    As[k4 * 4 + 1][m] = r.a[p].y;
+    As[k4 * 4 + 2][m] = r.a[p].z;
+    As[k4 * 4 + 3][m] = r.a[p].w;
+  }
+}
+
+// B tile direct global->LDS DMA: the row-major [BK][BN] image is lane-linear
+// for each wave (64 lanes x 16 B = two 128-float rows), so glds needs no
+// swizzle (guide §5 rule 21: linear dest, per-lane SOURCE address).
+__device__ __forceinline__ void glds_b(const float* B, int N, int block_col,
+                                       int k0, int tid, float (*BsBuf)[BN]) {
+  const int wavebase = (tid >> 6) * 2;  // first kk row this wave fills
+  const int n4 = tid & 31;
+  const int kk0 = tid >> 5;
+  for (int p = 0; p < 4; ++p) {
+    const int kk = kk0 + p * 8;
+    const float* src = B + (long)(k0 + kk) * N + block_col + n4 * 4;
+    auto* lbase = (__attribute__((address_space(3))) unsigned int*)
+        &BsBuf[wavebase + p * 8][0];
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src, lbase,
+        16, 0, 0);
+  }
+}
+
 template <int VARIANT>
 __global__ __launch_bounds__(256) void gemm_v(const float* __restrict__ A,
                                               const float* __restrict__ B,
@@ -75,8 +115,15 @@ __global__ __launch_bounds__(256) void gemm_v(const float* __restrict__ A,
 
   for (int kt = 0; kt < ktiles; ++kt) {
     __syncthreads();
-    if (kt + 1 < ktiles)
-      stage_load(A, B, K, N, block_row, block_col, (kt + 1) * BK, tid, regs);
+    if (kt + 1 < ktiles) {
+      if constexpr (VARIANT == 4) {
+        stage_load_a(A, K, block_row, (kt + 1) * BK, tid, regs);
+        glds_b(B, N, block_col, (kt + 1) * BK, tid, Bs[cur ^ 1]);
+      } else {
+        stage_load(A, B, K, N, block_row, block_col, (kt + 1) * BK, tid,
+                   regs);
+      }
+    }
 
     if constexpr (VARIANT == 0) {
       // baseline: load-then-mfma per 2-wide k step
@@ -124,6 +171,18 @@ __global__ __launch_bounds__(256) void gemm_v(const float* __restrict__ A,
         acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.y, b01.x, acc[1][0], 0, 0, 0);
         acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.y, b01.y, acc[1][1], 0, 0, 0);
       }
+    } else if constexpr (VARIANT == 4) {
+      // same inner loop as v0; v4 differs only in staging (B via glds below)
+      for (int kk = 0; kk < BK; kk += 2) {
+        const float a0 = As[cur][kk + kh][wr + l31];
+        const float a1 = As[cur][kk + kh][wr + 32 + l31];
+        const float b0 = Bs[cur][kk + kh][wc + l31];
+        const float b1 = Bs[cur][kk + kh][wc + 32 + l31];
+        acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+      }
     } else if constexpr (VARIANT == 3) {
       // setprio around the MFMA cluster (T5)
       for (int kk = 0; kk < BK; kk += 2) {
@@ -140,7 +199,12 @@ __global__ __launch_bounds__(256) void gemm_v(const float* __restrict__ A,
       }
     }
 
-    if (kt + 1 < ktiles) stage_write(As[cur ^ 1], Bs[cur ^ 1], tid, regs);
+    if (kt + 1 < ktiles) {
+      if constexpr (VARIANT == 4)
+        stage_write_a(As[cur ^ 1], tid, regs);
+      else
+        stage_write(As[cur ^ 1], Bs[cur ^ 1], tid, regs);
+    }
     cur ^= 1;
   }
 
@@ -309,10 +373,21 @@ int main(int argc, char** argv) {
     for (long i = 0; i < (long)Nsz * Nsz; ++i) bad += (h0[i] != h1[i]);
     printf("bigcheck: %ld mismatches\n", bad);
   }
+  {  // bit-check v4 (glds B staging) vs v0
+    std::vector<float> h0((long)Nsz * Nsz), h1((long)Nsz * Nsz);
+    dim3 g0(Nsz / BM, Nsz / BN);
+    hipLaunchKernelGGL((gemm_v<0>), g0, dim3(256), 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+    HIP_CHECK(hipMemcpy(h0.data(), dC, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
+    hipLaunchKernelGGL((gemm_v<4>), g0, dim3(256), 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+    HIP_CHECK(hipMemcpy(h1.data(), dC, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
+    long bad = 0;
+    for (long i = 0; i < (long)Nsz * Nsz; ++i) bad += (h0[i] != h1[i]);
+    printf("gldscheck: %ld mismatches\n", bad);
+  }
   for (int round = 0; round < rounds; ++round) {
-    printf("round %d: v0=%6.1f v1=%6.1f v3=%6.1f big=%6.1f TF\n", round,
+    printf("round %d: v0=%6.1f v1=%6.1f v4glds=%6.1f big=%6.1f TF\n", round,
            bench<0>(dA, dB, dC, Nsz, reps), bench<1>(dA, dB, dC, Nsz, reps),
-           bench<3>(dA, dB, dC, Nsz, reps), bench_big(dA, dB, dC, Nsz, reps));
+           bench<4>(dA, dB, dC, Nsz, reps), bench_big(dA, dB, dC, Nsz, reps));
   }
   return 0;
 }
