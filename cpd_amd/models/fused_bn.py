@@ -75,13 +75,15 @@ class FusedBNReLU(nn.Module):
         return F.relu(y) if self.relu else y
 
     def forward(self, x, residual=None):
+        # buffer update happens on every training path (fused or eager) so
+        # the counter stays consistent with nn.BatchNorm2d semantics
+        if self.training:
+            self.num_batches_tracked += 1
         use_fused = (self.training and x.is_cuda
                      and x.dtype == torch.float32
                      and (x.shape[2] * x.shape[3]) % 4 == 0)
         if not use_fused:
             return self._eager(x, residual)
-        if self.training:
-            self.num_batches_tracked += 1
         res = residual.contiguous() if residual is not None else None
         return _FusedBNFn.apply(x.contiguous(), self.weight, self.bias,
                                 self.running_mean, self.running_var,

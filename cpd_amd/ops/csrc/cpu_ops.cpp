@@ -139,7 +139,7 @@ Tensor seg_max_exp(const Tensor& flat, const Tensor& offsets,
       float m = 0.0f;
       for (int64_t i = ofs[s]; i < ofs[s + 1]; ++i)
         m = std::max(m, std::fabs(x[i]));
-      o[s] = cpd::ceil_log2_abs(m * (float)world_size);
+      o[s] = cpd::aps_max_exp(m, (int)world_size);
     }
   });
   return out;

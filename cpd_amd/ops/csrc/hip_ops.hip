@@ -285,7 +285,7 @@ __global__ void scale_quantize_aligned_kernel(float4* __restrict__ x, long n4,
 __global__ void maxabs_to_exp_kernel(const unsigned* __restrict__ bits,
                                      float* __restrict__ out, int S, int W) {
   const int s = blockIdx.x * TPB + threadIdx.x;
-  if (s < S) out[s] = ceil_log2_abs(bits_f32(bits[s]) * (float)W);
+  if (s < S) out[s] = aps_max_exp(bits_f32(bits[s]), W);
 }
 
 // flat[i] = Q(flat[i] * 2^shift[seg(i)])

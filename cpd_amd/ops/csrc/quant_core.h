@@ -145,4 +145,22 @@ CPD_HD float ceil_log2_abs(float x) {
   return (float)(e - 127 + (m != 0 ? 1 : 0));
 }
 
+// APS per-segment exponent: ceil(log2(max|g| * W)).  The fp32 product can
+// spuriously overflow to Inf when max|g| is finite but near FLT_MAX/W, which
+// would return 129 and silently crush the whole segment toward zero (ADVICE
+// r01); fall back to exact integer math with the conservative upper bound
+// ceil(log2 m) + ceil(log2 W) in that case.  A genuinely infinite max|g|
+// still yields 129 (Inf elements propagate through quantize/sum/unscale).
+CPD_HD float aps_max_exp(float maxabs, int world_size) {
+  const float p = maxabs * (float)world_size;
+  const uint32_t pu = f32_bits(p) & 0x7FFFFFFFu;
+  const uint32_t mu = f32_bits(maxabs) & 0x7FFFFFFFu;
+  if (pu == 0x7F800000u && mu < 0x7F800000u) {
+    int lw = 0;
+    while ((1 << lw) < world_size) ++lw;  // ceil(log2(W)), W >= 1
+    return ceil_log2_abs(maxabs) + (float)lw;
+  }
+  return ceil_log2_abs(p);
+}
+
 }  // namespace cpd

@@ -24,6 +24,8 @@ Semantics contract (documented divergence, SURVEY.md §7 hard-part 2):
     guaranteed per-chunk up to that rotation (validated by tests against
     the rotated sequential oracle).
 """
+import os
+
 import torch
 import torch.distributed as dist
 
@@ -35,9 +37,23 @@ def _wire_dtype(flat, man_bits, wire):
     # (exp<=8, man<=7) grid (e.g. after the APS scale+quantize pass), so it is
     # opt-in: callers that know the grads are on-grid pass wire="bf16"
     # (sum_gradients does).  Default is the always-exact f32 wire.
+    if wire == "bf16" and man_bits > 7:
+        raise ValueError(
+            f"wire='bf16' cannot represent man={man_bits} > 7 grids exactly; "
+            "use the f32 wire")
     if wire is not None:
         return {"bf16": torch.bfloat16, "f32": torch.float32}[wire]
     return torch.float32
+
+
+def _check_on_grid(flat, wdt):
+    # CPD_DEBUG_WIRE=1: verify the caller's on-grid claim — the bf16
+    # downcast must round-trip exactly (one pass over the bucket; debug only)
+    if wdt is torch.bfloat16 and os.environ.get("CPD_DEBUG_WIRE") == "1":
+        if not torch.equal(flat.to(wdt).float(), flat):
+            raise RuntimeError(
+                "wire='bf16' requested but data is not on a bf16 grid "
+                "(caller must APS-scale+quantize to man<=7 first)")
 
 
 def _qadd(acc, inc, man, exp):
@@ -98,6 +114,7 @@ def ring_lp_all_reduce_(flat, grad_exp, grad_man, use_kahan=False, group=None,
     n = flat.numel()
 
     wdt = _wire_dtype(flat, man, wire)
+    _check_on_grid(flat, wdt)
     chunk = (n + W - 1) // W
     padded = chunk * W
     if wdt == torch.float32 and padded == n:

@@ -160,8 +160,13 @@ class DistributedSampler(Sampler):
         self.rank = rank
         self.round_up = round_up
         self.epoch = 0
-        self.num_samples = int(math.ceil(len(dataset) / world_size))
-        self.total_size = self.num_samples * world_size
+        if round_up:
+            self.num_samples = int(math.ceil(len(dataset) / world_size))
+        else:
+            # actual per-rank slice length of indices[rank::world_size]
+            self.num_samples = (len(dataset) - rank + world_size - 1) \
+                // world_size
+        self.total_size = int(math.ceil(len(dataset) / world_size)) * world_size
 
     def set_epoch(self, epoch):
         self.epoch = epoch
