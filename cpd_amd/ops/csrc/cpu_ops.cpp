@@ -5,6 +5,7 @@
 // GPU tests compare against (same quant_core.h compiled for both targets).
 #include <torch/extension.h>
 #include <ATen/Parallel.h>
+#include <atomic>
 #include <cstring>
 #include <vector>
 
@@ -234,9 +235,36 @@ Tensor ceil_log2(const Tensor& x) {
   return out;
 }
 
+// Bit-equivalence scan: cast_fp_fast vs cast_fp over the uint32 bit-pattern
+// space with the given stride (stride=1 = the full 2^32 sweep).  Returns the
+// first mismatching bit pattern as int64, or -1 when none.  NaN payloads
+// compare as passthrough-identical bitwise (both return x unchanged).
+int64_t cast_fast_equiv_scan(int64_t man_bits, int64_t exp_bits,
+                             int64_t stride, int64_t offset) {
+  const int64_t total = (int64_t)1 << 32;
+  std::atomic<int64_t> bad{-1};
+  at::parallel_for(0, (total - offset + stride - 1) / stride, 1 << 16,
+                   [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e && bad.load(std::memory_order_relaxed) < 0;
+         ++i) {
+      const uint32_t u = (uint32_t)(offset + i * stride);
+      const float x = cpd::bits_f32(u);
+      const float a = cpd::cast_fp(x, (int)man_bits, (int)exp_bits);
+      const float b2 = cpd::cast_fp_fast(x, (int)man_bits, (int)exp_bits);
+      if (cpd::f32_bits(a) != cpd::f32_bits(b2)) {
+        int64_t expect = -1;
+        bad.compare_exchange_strong(expect, (int64_t)u);
+      }
+    }
+  });
+  return bad.load();
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("cast_fast_equiv_scan", &cast_fast_equiv_scan,
+        "first bit pattern where cast_fp_fast != cast_fp, else -1");
   m.def("quantize", &quantize, "FP32 -> (exp,man) grid, out-of-place");
   m.def("quantize_", &quantize_, "FP32 -> (exp,man) grid, in-place");
   m.def("qadd_", &qadd_, "acc = Q(acc + inc)");

@@ -254,8 +254,10 @@ __global__ __launch_bounds__(256) void quant_gemm_kernel(
       b[1] = Bs[kk][tx * 2 + 1];
       for (int i = 0; i < 2; ++i)
         for (int j = 0; j < 2; ++j) {
-          const float prod = cast_fp(a[i] * b[j], man, exp);
-          kahan_qstep(acc[i][j], comp[i][j], prod, man, exp);
+          // cast_fp_fast: ~20-VALU float-pipeline cast, bit-identical to
+          // cast_fp (exhaustive 2^32 sweep on host + device, r02)
+          const float prod = cast_fp_fast(a[i] * b[j], man, exp);
+          kahan_qstep_fast(acc[i][j], comp[i][j], prod, man, exp);
         }
     }
     __syncthreads();

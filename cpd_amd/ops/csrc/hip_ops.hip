@@ -501,6 +501,34 @@ Tensor ceil_log2(const Tensor& x) {
   return out;
 }
 
+// Device-side bit-equivalence scan of cast_fp_fast (the v_frexp/v_ldexp/
+// v_rndne formulation used in the quant_gemm inner loop) against cast_fp
+// over the 2^32 bit-pattern space.  Returns the first (lowest recorded)
+// mismatching pattern, or -1 when equal.  stride=1 = exhaustive.
+__global__ void cast_equiv_kernel(long n, long stride, int man, int exp,
+                                  unsigned long long* bad) {
+  const long gs = (long)gridDim.x * TPB;
+  for (long i = (long)blockIdx.x * TPB + threadIdx.x; i < n; i += gs) {
+    const unsigned u = (unsigned)(i * stride);
+    const float x = bits_f32(u);
+    const float a = cast_fp(x, man, exp);
+    const float b = cast_fp_fast(x, man, exp);
+    if (f32_bits(a) != f32_bits(b)) atomicMin(bad, (unsigned long long)u);
+  }
+}
+
+int64_t cast_fast_equiv_scan(int64_t man, int64_t exp, int64_t stride) {
+  const long total = ((long)1 << 32) / stride;
+  auto opts = at::TensorOptions().dtype(at::kLong).device(at::kCUDA);
+  Tensor bad = at::full({1}, -1, opts);  // 0xFFFF... as unsigned max
+  hipLaunchKernelGGL(cast_equiv_kernel, dim3(4096), dim3(TPB), 0,
+                     c10::hip::getCurrentHIPStream().stream(), total, stride,
+                     (int)man, (int)exp,
+                     (unsigned long long*)bad.data_ptr<int64_t>());
+  const int64_t v = bad.cpu().item<int64_t>();
+  return v;  // -1 (all ones) when no mismatch
+}
+
 }  // namespace
 
 // gemm_hip.hip provides these at namespace scope
@@ -521,6 +549,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_quantize_", &scale_quantize_);
   m.def("seg_scale_", &seg_scale_);
   m.def("ceil_log2", &ceil_log2);
+  m.def("cast_fast_equiv_scan", &cast_fast_equiv_scan);
   m.def("quant_gemm", &cpd_quant_gemm_hip);
   m.def("gemm_f32", &cpd_gemm_f32_hip);
 }

@@ -113,6 +113,40 @@ CPD_HD float cast_fp(float x, int man_bits, int exp_bits) {
   return (au - 1u >= 0x7F7FFFFFu) ? x : res;  // +-0 / Inf / NaN passthrough
 }
 
+// Float-pipeline formulation of cast_fp — bit-identical semantics, ~20 VALU
+// ops instead of ~35 (the integer path's shift/mask/reassemble sequence).
+// On gfx950 each step maps to one full-rate instruction: v_frexp_exp_i32_f32,
+// v_ldexp_f32, v_trunc_f32, v_rndne_f32.  Derivation (all steps exact):
+//   ax = m24 * 2^(e-1-23), m24 in [2^23, 2^24)      (frexp: ax = m*2^e, m in [.5,1))
+//   t1 = trunc(ax * 2^k1) = m24 >> shift            (sticky-discarding pre-shift;
+//        k1 = 23-(e-1)-shift; shift >= 24 flushes to 0 exactly like the int path)
+//   r  = rndne(t1 * 2^-drop)                        (RNE at man_bits)
+//   mag = r * 2^(drop-k1)                           (== scale_pow2(man, out_e-23)
+//        in BOTH the normal and target-subnormal branches — see k1 algebra)
+// Specials (overflow saturate, fp32-subnormal flush, +-0/Inf/NaN passthrough)
+// are the same selects as cast_fp.  Used in the quant_gemm inner loop where
+// the cast IS the cost model; bit-equality with cast_fp is tested on CPU over
+// random bit patterns and on GPU (tests/test_quantize.py, test_gpu_numerics).
+CPD_HD float cast_fp_fast(float x, int man_bits, int exp_bits) {
+  const uint32_t u = f32_bits(x);
+  const uint32_t au = u & 0x7FFFFFFFu;
+  const float ax = bits_f32(au);
+  int e;
+  (void)frexpf(ax, &e);  // device: v_frexp_exp_i32_f32
+  const int bias = (1 << (exp_bits - 1)) - 1;
+  const int new_e = (e - 1) + bias;
+  const int shift = new_e > 0 ? 0 : 1 - new_e;  // no upper clamp: the float
+  const int k1 = 23 - (e - 1) - shift;          // path flushes naturally
+  const float t1 = truncf(ldexpf(ax, k1));
+  const int drop = 23 - man_bits;
+  const float r = rintf(ldexpf(t1, -drop));  // v_rndne_f32 (RNE)
+  float mag = ldexpf(r, drop - k1);
+  mag = new_e >= (1 << exp_bits) - 1 ? bits_f32(0x7F800000u) : mag;  // ovf
+  float res = bits_f32(f32_bits(mag) | (u & 0x80000000u));
+  res = au < 0x00800000u ? 0.0f : res;        // fp32 subnormal flush
+  return (au - 1u >= 0x7F7FFFFFu) ? x : res;  // +-0 / Inf / NaN passthrough
+}
+
 // One step of (exp,man)-rounded Kahan compensated accumulation:
 //   y = Q(inc - c); t = Q(acc + y); c = Q(Q(t - acc) - y); acc = t
 // Every intermediate is rounded, matching the reference's gradient-sum and
@@ -122,6 +156,16 @@ CPD_HD void kahan_qstep(float& acc, float& c, float inc, int man_bits,
   const float y = cast_fp(inc - c, man_bits, exp_bits);
   const float t = cast_fp(acc + y, man_bits, exp_bits);
   c = cast_fp(cast_fp(t - acc, man_bits, exp_bits) - y, man_bits, exp_bits);
+  acc = t;
+}
+
+// Same step through cast_fp_fast (bit-identical; the quant_gemm hot loop).
+CPD_HD void kahan_qstep_fast(float& acc, float& c, float inc, int man_bits,
+                             int exp_bits) {
+  const float y = cast_fp_fast(inc - c, man_bits, exp_bits);
+  const float t = cast_fp_fast(acc + y, man_bits, exp_bits);
+  c = cast_fp_fast(cast_fp_fast(t - acc, man_bits, exp_bits) - y, man_bits,
+                   exp_bits);
   acc = t;
 }
 

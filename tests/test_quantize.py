@@ -103,3 +103,16 @@ def test_quantizer_autograd():
     q_id = quantizer()
     x2 = torch.tensor([1e-45, 1.0], requires_grad=True)
     assert torch.equal(q_id(x2).detach(), x2.detach())
+
+
+def test_cast_fast_equiv_host_sampled():
+    """cast_fp_fast vs cast_fp on host: strided scan over the full bit-pattern
+    space (the full 2^32 sweep passed for e4m3/e5m2/e3m0/e8m23/e8m7 during
+    development; stride keeps CI fast while still crossing every exponent)."""
+    from cpd_amd import ops as _ops
+    cpu = _ops.cpu_ext()
+    for (exp, man) in [(4, 3), (5, 2), (8, 23), (3, 0), (6, 9)]:
+        bad = cpu.cast_fast_equiv_scan(man, exp, 65537, 0)
+        assert bad == -1, f"e{exp}m{man}: {bad:#x}"
+        bad = cpu.cast_fast_equiv_scan(man, exp, 65537, 12345)
+        assert bad == -1, f"e{exp}m{man} offset: {bad:#x}"

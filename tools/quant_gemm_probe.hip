@@ -1,4 +1,8 @@
 // Variant probe for the (exp,man)-Kahan-accumulator GEMM (VALU kernel).
+// Round 2: cast_fp_fast (v_frexp/v_ldexp/v_trunc/v_rndne formulation, ~20
+// VALU vs ~35) against the integer-path baseline, plus tile/K variants.
+// Every variant preserves the strict per-output-element K order, so all
+// results must be BIT-IDENTICAL — checked before timing.
 // Build: hipcc --offload-arch=gfx950 -O3 -Wno-unused-value tools/quant_gemm_probe.hip -o tools/quant_gemm_probe
 #include <hip/hip_runtime.h>
 #include <cstdio>
@@ -12,94 +16,16 @@
   printf("HIP error %s @%d\n", hipGetErrorString(e), __LINE__); exit(1);} } while (0)
 
 using namespace cpd;
-constexpr int QBM = 64, QBN = 64, QBK = 16;
 
-template <int VARIANT>
-__global__ __launch_bounds__(256) void quant_gemm_v(
+// ---------------------------------------------------------------------------
+// 32x32 tile, 16x16 threads, 2x2 outputs/lane.  FAST selects cast_fp_fast.
+// TK = K-tile depth (barrier period).
+// ---------------------------------------------------------------------------
+template <bool FAST, int TK>
+__global__ __launch_bounds__(256) void qg32(
     const float* __restrict__ A, const float* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, int man, int exp) {
-  __shared__ float As[QBK][QBM + 1];
-  __shared__ float Bs[QBK][QBN];
-  const int tx = threadIdx.x & 15;
-  const int ty = threadIdx.x >> 4;
-  const int row0 = blockIdx.x * QBM + ty * 4;
-  const int col0 = blockIdx.y * QBN + tx * 4;
-  float acc[4][4] = {};
-  float comp[4][4] = {};
-  const int ktiles = (K + QBK - 1) / QBK;
-  for (int kt = 0; kt < ktiles; ++kt) {
-    const int k0 = kt * QBK;
-    {
-      const int k = threadIdx.x & 15;
-      const int m0 = threadIdx.x >> 4;
-      for (int p = 0; p < 4; ++p) {
-        const int m = m0 + p * 16;
-        const int gm = blockIdx.x * QBM + m;
-        As[k][m] = (gm < M && k0 + k < K) ? A[(long)gm * K + k0 + k] : 0.0f;
-      }
-      const int n = threadIdx.x & 63;
-      const int kk0 = threadIdx.x >> 6;
-      for (int p = 0; p < 4; ++p) {
-        const int kk = kk0 + p * 4;
-        const int gn = blockIdx.y * QBN + n;
-        Bs[kk][n] = (k0 + kk < K && gn < N) ? B[(long)(k0 + kk) * N + gn] : 0.0f;
-      }
-    }
-    __syncthreads();
-    const int klim = min(QBK, K - k0);
-    for (int kk = 0; kk < klim; ++kk) {
-      float a[4], b[4];
-#pragma unroll
-      for (int i = 0; i < 4; ++i) a[i] = As[kk][ty * 4 + i];
-#pragma unroll
-      for (int j = 0; j < 4; ++j) b[j] = Bs[kk][tx * 4 + j];
-      if constexpr (VARIANT == 0) {
-        // nested per-output kahan (current production structure)
-#pragma unroll
-        for (int i = 0; i < 4; ++i)
-#pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            const float prod = cast_fp(a[i] * b[j], man, exp);
-            kahan_qstep(acc[i][j], comp[i][j], prod, man, exp);
-          }
-      } else if constexpr (VARIANT == 1) {
-        // stage-batched: run each rounding stage across all 16 chains
-        float prod[16], yv[16], tv[16];
-#pragma unroll
-        for (int i = 0; i < 4; ++i)
-#pragma unroll
-          for (int j = 0; j < 4; ++j)
-            prod[i * 4 + j] = cast_fp(a[i] * b[j], man, exp);
-#pragma unroll
-        for (int q = 0; q < 16; ++q)
-          yv[q] = cast_fp(prod[q] - comp[q >> 2][q & 3], man, exp);
-#pragma unroll
-        for (int q = 0; q < 16; ++q)
-          tv[q] = cast_fp(acc[q >> 2][q & 3] + yv[q], man, exp);
-#pragma unroll
-        for (int q = 0; q < 16; ++q) {
-          comp[q >> 2][q & 3] = cast_fp(
-              cast_fp(tv[q] - acc[q >> 2][q & 3], man, exp) - yv[q], man, exp);
-          acc[q >> 2][q & 3] = tv[q];
-        }
-      }
-    }
-    __syncthreads();
-  }
-  for (int i = 0; i < 4; ++i) {
-    if (row0 + i >= M) break;
-    for (int j = 0; j < 4; ++j)
-      if (col0 + j < N) C[(long)(row0 + i) * N + col0 + j] = acc[i][j];
-  }
-}
-
-// 32x32 tile, 2x2 per thread: 4x the resident waves (the 64x64 tile yields
-// only (N/64)^2 blocks = 1 wave/SIMD at 1024^3 — a latency-bound VALU kernel
-// then eats every cast-chain dependency stall).
-__global__ __launch_bounds__(256) void quant_gemm_32(
-    const float* __restrict__ A, const float* __restrict__ B,
-    float* __restrict__ C, int M, int N, int K, int man, int exp) {
-  constexpr int TB = 32, TK = 16;
+  constexpr int TB = 32;
   __shared__ float As[TK][TB + 1];
   __shared__ float Bs[TK][TB];
   const int tx = threadIdx.x & 15;
@@ -112,17 +38,14 @@ __global__ __launch_bounds__(256) void quant_gemm_32(
   for (int kt = 0; kt < ktiles; ++kt) {
     const int k0 = kt * TK;
     {
-      const int k = threadIdx.x & 15;       // A: [32][16] transposed, 2/thread
-      const int m0 = threadIdx.x >> 4;
-      for (int p = 0; p < 2; ++p) {
-        const int m = m0 + p * 16;
+      // A: [TB][TK] transposed; TB*TK elems over 256 threads
+      for (int idx = threadIdx.x; idx < TB * TK; idx += 256) {
+        const int k = idx % TK, m = idx / TK;
         const int gm = blockIdx.x * TB + m;
         As[k][m] = (gm < M && k0 + k < K) ? A[(long)gm * K + k0 + k] : 0.0f;
       }
-      const int n = threadIdx.x & 31;       // B: [16][32], 2/thread
-      const int kk0 = threadIdx.x >> 5;
-      for (int p = 0; p < 2; ++p) {
-        const int kk = kk0 + p * 8;
+      for (int idx = threadIdx.x; idx < TK * TB; idx += 256) {
+        const int n = idx % TB, kk = idx / TB;
         const int gn = blockIdx.y * TB + n;
         Bs[kk][n] = (k0 + kk < K && gn < N) ? B[(long)(k0 + kk) * N + gn] : 0.0f;
       }
@@ -137,8 +60,13 @@ __global__ __launch_bounds__(256) void quant_gemm_32(
       for (int i = 0; i < 2; ++i)
 #pragma unroll
         for (int j = 0; j < 2; ++j) {
-          const float prod = cast_fp(a[i] * b[j], man, exp);
-          kahan_qstep(acc[i][j], comp[i][j], prod, man, exp);
+          if constexpr (FAST) {
+            const float prod = cast_fp_fast(a[i] * b[j], man, exp);
+            kahan_qstep_fast(acc[i][j], comp[i][j], prod, man, exp);
+          } else {
+            const float prod = cast_fp(a[i] * b[j], man, exp);
+            kahan_qstep(acc[i][j], comp[i][j], prod, man, exp);
+          }
         }
     }
     __syncthreads();
@@ -150,87 +78,142 @@ __global__ __launch_bounds__(256) void quant_gemm_32(
   }
 }
 
-double bench32(const float* dA, const float* dB, float* dC, int Nsz,
-               int reps) {
-  dim3 grid((Nsz + 31) / 32, (Nsz + 31) / 32), block(256);
-  hipLaunchKernelGGL(quant_gemm_32, grid, block, 0, 0, dA, dB, dC, Nsz, Nsz,
-                     Nsz, 3, 4);
-  HIP_CHECK(hipDeviceSynchronize());
-  hipEvent_t t0, t1;
-  hipEventCreate(&t0); hipEventCreate(&t1);
-  hipEventRecord(t0);
-  for (int i = 0; i < reps; ++i)
-    hipLaunchKernelGGL(quant_gemm_32, grid, block, 0, 0, dA, dB, dC, Nsz, Nsz,
-                       Nsz, 3, 4);
-  hipEventRecord(t1);
-  HIP_CHECK(hipEventSynchronize(t1));
-  float ms; hipEventElapsedTime(&ms, t0, t1);
-  return 2.0 * Nsz * Nsz * (double)Nsz * reps / (ms * 1e-3) / 1e12;
+// 64x32 tile, 256 threads, 4x2 outputs/lane (8 chains: more ILP per lane,
+// half the blocks of the 32-tile).
+template <bool FAST>
+__global__ __launch_bounds__(256) void qg64x32(
+    const float* __restrict__ A, const float* __restrict__ B,
+    float* __restrict__ C, int M, int N, int K, int man, int exp) {
+  constexpr int TM = 64, TN = 32, TK = 16;
+  __shared__ float As[TK][TM + 1];
+  __shared__ float Bs[TK][TN];
+  const int tx = threadIdx.x & 15;   // 16 cols of lanes * 2 outputs
+  const int ty = threadIdx.x >> 4;   // 16 rows of lanes * 4 outputs
+  const int row0 = blockIdx.x * TM + ty * 4;
+  const int col0 = blockIdx.y * TN + tx * 2;
+  float acc[4][2] = {};
+  float comp[4][2] = {};
+  const int ktiles = (K + TK - 1) / TK;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    const int k0 = kt * TK;
+    for (int idx = threadIdx.x; idx < TM * TK; idx += 256) {
+      const int k = idx % TK, m = idx / TK;
+      const int gm = blockIdx.x * TM + m;
+      As[k][m] = (gm < M && k0 + k < K) ? A[(long)gm * K + k0 + k] : 0.0f;
+    }
+    for (int idx = threadIdx.x; idx < TK * TN; idx += 256) {
+      const int n = idx % TN, kk = idx / TN;
+      const int gn = blockIdx.y * TN + n;
+      Bs[kk][n] = (k0 + kk < K && gn < N) ? B[(long)(k0 + kk) * N + gn] : 0.0f;
+    }
+    __syncthreads();
+    const int klim = min(TK, K - k0);
+    for (int kk = 0; kk < klim; ++kk) {
+      float a[4], b[2];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) a[i] = As[kk][ty * 4 + i];
+      b[0] = Bs[kk][tx * 2]; b[1] = Bs[kk][tx * 2 + 1];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          if constexpr (FAST) {
+            const float prod = cast_fp_fast(a[i] * b[j], man, exp);
+            kahan_qstep_fast(acc[i][j], comp[i][j], prod, man, exp);
+          } else {
+            const float prod = cast_fp(a[i] * b[j], man, exp);
+            kahan_qstep(acc[i][j], comp[i][j], prod, man, exp);
+          }
+        }
+    }
+    __syncthreads();
+  }
+  for (int i = 0; i < 4; ++i) {
+    if (row0 + i >= M) break;
+    for (int j = 0; j < 2; ++j)
+      if (col0 + j < N) C[(long)(row0 + i) * N + col0 + j] = acc[i][j];
+  }
 }
 
-template <int V>
-double bench(const float* dA, const float* dB, float* dC, int Nsz, int reps) {
-  dim3 grid(Nsz / QBM, Nsz / QBN), block(256);
-  hipLaunchKernelGGL((quant_gemm_v<V>), grid, block, 0, 0, dA, dB, dC, Nsz,
-                     Nsz, Nsz, 3, 4);
-  HIP_CHECK(hipDeviceSynchronize());
-  hipEvent_t t0, t1;
-  hipEventCreate(&t0); hipEventCreate(&t1);
-  hipEventRecord(t0);
-  for (int i = 0; i < reps; ++i)
-    hipLaunchKernelGGL((quant_gemm_v<V>), grid, block, 0, 0, dA, dB, dC, Nsz,
-                       Nsz, Nsz, 3, 4);
-  hipEventRecord(t1);
-  HIP_CHECK(hipEventSynchronize(t1));
-  float ms; hipEventElapsedTime(&ms, t0, t1);
-  return 2.0 * Nsz * Nsz * (double)Nsz * reps / (ms * 1e-3) / 1e12;
+// -------------------------- harness ----------------------------------------
+
+struct Variant {
+  const char* name;
+  void (*launch)(const float*, const float*, float*, int, int);
+};
+
+template <bool FAST, int TK>
+void launch32(const float* dA, const float* dB, float* dC, int Nsz, int me) {
+  dim3 grid((Nsz + 31) / 32, (Nsz + 31) / 32);
+  hipLaunchKernelGGL((qg32<FAST, TK>), grid, dim3(256), 0, 0, dA, dB, dC,
+                     Nsz, Nsz, Nsz, me & 0xff, me >> 8);
+}
+
+template <bool FAST>
+void launch64x32(const float* dA, const float* dB, float* dC, int Nsz, int me) {
+  dim3 grid((Nsz + 63) / 64, (Nsz + 31) / 32);
+  hipLaunchKernelGGL((qg64x32<FAST>), grid, dim3(256), 0, 0, dA, dB, dC,
+                     Nsz, Nsz, Nsz, me & 0xff, me >> 8);
 }
 
 int main(int argc, char** argv) {
   const int Nsz = argc > 1 ? atoi(argv[1]) : 1024;
   const int reps = argc > 2 ? atoi(argv[2]) : 3;
+  const int man = argc > 3 ? atoi(argv[3]) : 3;
+  const int exp = argc > 4 ? atoi(argv[4]) : 4;
+  const int me = man | (exp << 8);
   std::vector<float> hA((long)Nsz * Nsz), hB((long)Nsz * Nsz);
   srand(1);
   for (auto& v : hA) v = (rand() / (float)RAND_MAX) * 2 - 1;
   for (auto& v : hB) v = (rand() / (float)RAND_MAX) * 2 - 1;
-  float *dA, *dB, *dC0, *dC1;
+  float *dA, *dB, *dRef, *dC;
   HIP_CHECK(hipMalloc(&dA, (long)Nsz * Nsz * 4));
   HIP_CHECK(hipMalloc(&dB, (long)Nsz * Nsz * 4));
-  HIP_CHECK(hipMalloc(&dC0, (long)Nsz * Nsz * 4));
-  HIP_CHECK(hipMalloc(&dC1, (long)Nsz * Nsz * 4));
+  HIP_CHECK(hipMalloc(&dRef, (long)Nsz * Nsz * 4));
+  HIP_CHECK(hipMalloc(&dC, (long)Nsz * Nsz * 4));
   HIP_CHECK(hipMemcpy(dA, hA.data(), (long)Nsz * Nsz * 4, hipMemcpyHostToDevice));
   HIP_CHECK(hipMemcpy(dB, hB.data(), (long)Nsz * Nsz * 4, hipMemcpyHostToDevice));
-  // bit-equality check v1 vs v0 (same rounding order is required)
-  {
-    dim3 grid(Nsz / QBM, Nsz / QBN), block(256);
-    hipLaunchKernelGGL((quant_gemm_v<0>), grid, block, 0, 0, dA, dB, dC0,
-                       Nsz, Nsz, Nsz, 3, 4);
-    hipLaunchKernelGGL((quant_gemm_v<1>), grid, block, 0, 0, dA, dB, dC1,
-                       Nsz, Nsz, Nsz, 3, 4);
+
+  struct V { const char* name; void (*fn)(const float*, const float*, float*, int, int); };
+  V vs[] = {
+      {"v32 base (int cast)  ", launch32<false, 16>},
+      {"v32 fast cast        ", launch32<true, 16>},
+      {"v32 fast cast TK=32  ", launch32<true, 32>},
+      {"v64x32 fast (8 chain)", launch64x32<true>},
+  };
+  const int NV = sizeof(vs) / sizeof(vs[0]);
+
+  // bit-identity of every variant vs the baseline
+  vs[0].fn(dA, dB, dRef, Nsz, me);
+  HIP_CHECK(hipDeviceSynchronize());
+  std::vector<float> h0((long)Nsz * Nsz), h1((long)Nsz * Nsz);
+  HIP_CHECK(hipMemcpy(h0.data(), dRef, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
+  for (int v = 1; v < NV; ++v) {
+    vs[v].fn(dA, dB, dC, Nsz, me);
     HIP_CHECK(hipDeviceSynchronize());
-    std::vector<float> h0((long)Nsz * Nsz), h1((long)Nsz * Nsz);
-    HIP_CHECK(hipMemcpy(h0.data(), dC0, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
-    HIP_CHECK(hipMemcpy(h1.data(), dC1, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(h1.data(), dC, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
+    long bad = -1;
     for (long i = 0; i < (long)Nsz * Nsz; ++i)
-      if (h0[i] != h1[i]) { printf("BITCHECK FAIL at %ld\n", i); return 1; }
-    printf("bitcheck ok\n");
+      if (h0[i] != h1[i]) { bad = i; break; }
+    printf("%s bitcheck %s\n", vs[v].name, bad < 0 ? "ok" : "FAIL");
+    if (bad >= 0) return 1;
   }
-  // bit-check the 32-tile variant (same K order per output element)
-  {
-    dim3 g32((Nsz + 31) / 32, (Nsz + 31) / 32);
-    hipLaunchKernelGGL(quant_gemm_32, g32, dim3(256), 0, 0, dA, dB, dC1,
-                       Nsz, Nsz, Nsz, 3, 4);
-    HIP_CHECK(hipDeviceSynchronize());
-    std::vector<float> h0((long)Nsz * Nsz), h1((long)Nsz * Nsz);
-    HIP_CHECK(hipMemcpy(h0.data(), dC0, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
-    HIP_CHECK(hipMemcpy(h1.data(), dC1, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
-    for (long i = 0; i < (long)Nsz * Nsz; ++i)
-      if (h0[i] != h1[i]) { printf("32TILE BITCHECK FAIL at %ld\n", i); return 1; }
-    printf("32-tile bitcheck ok\n");
+
+  for (int r = 0; r < 3; ++r) {
+    for (int v = 0; v < NV; ++v) {
+      vs[v].fn(dA, dB, dC, Nsz, me);  // warm
+      HIP_CHECK(hipDeviceSynchronize());
+      hipEvent_t t0, t1;
+      hipEventCreate(&t0); hipEventCreate(&t1);
+      hipEventRecord(t0);
+      for (int i = 0; i < reps; ++i) vs[v].fn(dA, dB, dC, Nsz, me);
+      hipEventRecord(t1);
+      HIP_CHECK(hipEventSynchronize(t1));
+      float ms; hipEventElapsedTime(&ms, t0, t1);
+      const double tf = 2.0 * Nsz * Nsz * (double)Nsz * reps / (ms * 1e-3) / 1e12;
+      printf("round %d  %s %8.3f TF (e%dm%d)\n", r, vs[v].name, tf, exp, man);
+      hipEventDestroy(t0); hipEventDestroy(t1);
+    }
   }
-  for (int r = 0; r < 3; ++r)
-    printf("round %d: v0=%7.3f v1=%7.3f v32=%7.3f TF\n", r,
-           bench<0>(dA, dB, dC0, Nsz, reps), bench<1>(dA, dB, dC1, Nsz, reps),
-           bench32(dA, dB, dC1, Nsz, reps));
   return 0;
 }
