@@ -32,6 +32,35 @@ class SyntheticImages(Dataset):
         return self.images[j], self.labels[j]
 
 
+class ProceduralImages(Dataset):
+    """Deterministic LEARNABLE classification set (no network in the target
+    environment, so the APS-recovers-accuracy experiment — the reference's
+    core claim, README.md:153-154 — runs on procedurally generated data).
+
+    Each class owns `templates_per_class` fixed random templates; a sample is
+    a randomly scaled template plus unit Gaussian noise.  Fully determined by
+    (seed, index); train/val splits use different seeds but share templates,
+    so a model that learns the templates generalizes to the val split."""
+
+    def __init__(self, n=16384, shape=(3, 32, 32), num_classes=10, seed=0,
+                 templates_per_class=4, snr=0.35):
+        g = torch.Generator().manual_seed(777)  # templates shared by splits
+        self.templates = torch.randn(
+            (num_classes, templates_per_class) + tuple(shape), generator=g)
+        gs = torch.Generator().manual_seed(seed)
+        self.labels = torch.randint(0, num_classes, (n,), generator=gs)
+        v = torch.randint(0, templates_per_class, (n,), generator=gs)
+        s = 0.5 + torch.rand((n, 1, 1, 1), generator=gs)
+        noise = torch.randn((n,) + tuple(shape), generator=gs)
+        self.images = snr * s * self.templates[self.labels, v] + noise
+
+    def __len__(self):
+        return self.labels.numel()
+
+    def __getitem__(self, i):
+        return self.images[i], int(self.labels[i])
+
+
 def _load_cifar_batches(root):
     files = [f"data_batch_{i}" for i in range(1, 6)]
     xs, ys = [], []

@@ -55,6 +55,11 @@ def parse_args(argv=None):
                    help='sub-buckets for backward-overlapped reduction')
     p.add_argument('--synthetic', action='store_true',
                    help='synthetic CIFAR-shaped data (no dataset on disk)')
+    p.add_argument('--procedural', action='store_true',
+                   help='deterministic LEARNABLE procedural data (accuracy '
+                        'experiments without a dataset on disk)')
+    p.add_argument('--max_iter', default=0, type=int,
+                   help='override the epoch-derived iteration count (>0)')
     p.add_argument('--data-root', default='./data/cifar-10-batches-py')
     # config-file defaults (res18_cifar.yaml parity)
     p.add_argument('--arch', default='res_cifar')
@@ -118,7 +123,11 @@ def main(argv=None):
                        emulate_node=args.emulate_node, mode=args.mode,
                        overlap=args.overlap if args.emulate_node == 1 else 0)
 
-    if args.synthetic or not os.path.isdir(args.data_root):
+    if args.procedural:
+        from cpd_amd.data import ProceduralImages
+        train_set = ProceduralImages(16384, seed=0)
+        val_set = ProceduralImages(2048, seed=1)
+    elif args.synthetic or not os.path.isdir(args.data_root):
         if not args.synthetic and rank == 0:
             print(f'No CIFAR at {args.data_root}; using synthetic data.')
         train_set = SyntheticImages(50000)
@@ -128,7 +137,8 @@ def main(argv=None):
         val_set = CIFAR10(args.data_root, train=False, augment=False)
 
     denom = world_size * args.batch_size * args.emulate_node
-    max_iter = math.ceil(len(train_set) * args.max_epoch / denom)
+    max_iter = args.max_iter or math.ceil(
+        len(train_set) * args.max_epoch / denom)
     iter_per_epoch = math.ceil(len(train_set) / denom)
     last_iter = -1
 
