@@ -38,6 +38,9 @@ __device__ float block_reduce(float v, float* smem) {
 }
 
 // ---- forward pass 1: per (channel, slice) partial sum / sumsq -------------
+// The (n, hw) space is FLATTENED across the block so all 256 lanes stay
+// active even when HW < TPB*4 (late ResNet layers: HW=16 left 4 active
+// lanes/block in the r01 per-sample loop — 2-3x slower reduces).
 __global__ void bn_reduce_kernel(const float* __restrict__ x, int N, int C,
                                  long HW, int split,
                                  float* __restrict__ ws /* [C][split][2] */) {
@@ -45,13 +48,14 @@ __global__ void bn_reduce_kernel(const float* __restrict__ x, int N, int C,
   const int s = blockIdx.y;
   __shared__ float smem[TPB / 64];
   float sum = 0.f, sumsq = 0.f;
-  for (int n = s; n < N; n += split) {
-    const float* base = x + ((long)n * C + c) * HW;
-    for (long i = threadIdx.x * 4; i < HW; i += (long)TPB * 4) {
-      const float4 v = ld4(base + i);
-      sum += v.x + v.y + v.z + v.w;
-      sumsq += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
-    }
+  const long total4 = (long)N * HW / 4;  // HW % 4 == 0
+  for (long f4 = (long)s * TPB + threadIdx.x; f4 < total4;
+       f4 += (long)split * TPB) {
+    const long f = f4 * 4;
+    const long n = f / HW;
+    const float4 v = ld4(x + (n * C + c) * HW + (f - n * HW));
+    sum += v.x + v.y + v.z + v.w;
+    sumsq += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
   }
   const float bs = block_reduce(sum, smem);
   __syncthreads();
@@ -183,14 +187,18 @@ __global__ void bn_bwd_reduce_mask_kernel(const float* __restrict__ x,
   __shared__ float smem[TPB / 64];
   const float m = mean[c], is = invstd[c];
   float sd = 0.f, sdx = 0.f;
-  for (int n = s; n < N; n += split) {
-    const long off = ((long)n * C + c) * HW;
-    for (long i = threadIdx.x * 8; i < HW; i += (long)TPB * 8) {
-      float4 g0 = ld4(dy + off + i);
-      float4 g1 = ld4(dy + off + i + 4);
-      const float4 v0 = ld4(x + off + i);
-      const float4 v1 = ld4(x + off + i + 4);
-      const unsigned mk = mask[(off + i) / 8];
+  const long total8 = (long)N * HW / 8;  // flattened (n,hw): all lanes active
+  for (long f8 = (long)s * TPB + threadIdx.x; f8 < total8;
+       f8 += (long)split * TPB) {
+    const long f = f8 * 8;
+    const long n = f / HW;
+    const long off = (n * C + c) * HW + (f - n * HW);
+    {
+      float4 g0 = ld4(dy + off);
+      float4 g1 = ld4(dy + off + 4);
+      const float4 v0 = ld4(x + off);
+      const float4 v1 = ld4(x + off + 4);
+      const unsigned mk = mask[off / 8];
       g0.x = (mk & 1) ? g0.x : 0.f;
       g0.y = (mk & 2) ? g0.y : 0.f;
       g0.z = (mk & 4) ? g0.z : 0.f;
@@ -282,22 +290,24 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
   __shared__ float smem[TPB / 64];
   const float m = mean[c], is = invstd[c];
   float sd = 0.f, sdx = 0.f;
-  for (int n = s; n < N; n += split) {
-    const long off = ((long)n * C + c) * HW;
-    for (long i = threadIdx.x * 4; i < HW; i += (long)TPB * 4) {
-      float4 g = ld4(dy + off + i);
-      const float4 v = ld4(x + off + i);
-      if (y) {
-        const float4 yy = ld4(y + off + i);
-        g.x = yy.x > 0.f ? g.x : 0.f;
-        g.y = yy.y > 0.f ? g.y : 0.f;
-        g.z = yy.z > 0.f ? g.z : 0.f;
-        g.w = yy.w > 0.f ? g.w : 0.f;
-      }
-      sd += g.x + g.y + g.z + g.w;
-      sdx += g.x * (v.x - m) + g.y * (v.y - m) + g.z * (v.z - m) +
-             g.w * (v.w - m);
+  const long total4 = (long)N * HW / 4;  // flattened (n,hw): all lanes active
+  for (long f4 = (long)s * TPB + threadIdx.x; f4 < total4;
+       f4 += (long)split * TPB) {
+    const long f = f4 * 4;
+    const long n = f / HW;
+    const long off = (n * C + c) * HW + (f - n * HW);
+    float4 g = ld4(dy + off);
+    const float4 v = ld4(x + off);
+    if (y) {
+      const float4 yy = ld4(y + off);
+      g.x = yy.x > 0.f ? g.x : 0.f;
+      g.y = yy.y > 0.f ? g.y : 0.f;
+      g.z = yy.z > 0.f ? g.z : 0.f;
+      g.w = yy.w > 0.f ? g.w : 0.f;
     }
+    sd += g.x + g.y + g.z + g.w;
+    sdx += g.x * (v.x - m) + g.y * (v.y - m) + g.z * (v.z - m) +
+           g.w * (v.w - m);
   }
   const float bs = block_reduce(sd, smem);
   __syncthreads();

@@ -259,6 +259,12 @@ def test_hip_graph_step_matches_eager():
         dm = DistModule(model)
         opt = torch.optim.SGD([{"params": model.parameters()}], lr=0.1,
                               momentum=0.9, weight_decay=1e-4)
+        # pre-initialize momentum buffers: zeros gives the same first-step
+        # math as torch's lazy buf=grad.clone() init, and graph capture must
+        # not record the lazy init (it would replay buf=grad every step).
+        # bench.py gets this for free from its in-place warmup substeps.
+        for p in model.parameters():
+            opt.state[p] = {"momentum_buffer": torch.zeros_like(p)}
         step = LPTrainStep(dm, opt, grad_exp=4, grad_man=3, use_APS=True,
                            use_master=False)
         return model, step
