@@ -27,9 +27,12 @@ timeout 300 ./tools/quant_gemm_probe 2048 3 3 4 > gpurun_out/qgp2_2048.log 2>&1
 
 # clean kernel-trace profile of the eager NCHW bench steady state
 cd /tmp && export TMPDIR=/tmp && cd "$GRAFT_REPO_ROOT"
-timeout 600 rocprofv3 --kernel-trace --stats -d gpurun_out/prof_r02 -o r02 \
+timeout 600 rocprofv3 --kernel-trace --output-format rocpd \
+    -d gpurun_out/prof_r02 -o r02 \
     -- python bench.py --steps 12 --warmup 6 --no-hip-graph \
     > gpurun_out/prof_bench.log 2>&1
 echo "prof rc=$?"
-python tools/prof_summary.py gpurun_out/prof_r02 > gpurun_out/prof_steady_r02.txt 2>&1 || true
+DB=$(find gpurun_out/prof_r02 -name '*.db' | head -1)
+python tools/prof_summary.py "$DB" > gpurun_out/prof_steady_r02.txt 2>&1 || true
+rm -rf gpurun_out/prof_r02   # keep only the summary (merge-size budget)
 tail -1 gpurun_out/b_nchw_fused.log gpurun_out/b_nchw_eagerbn.log gpurun_out/b_cl_normal.log gpurun_out/b_cl_search.log
