@@ -96,7 +96,9 @@ def main():
         os.environ.get("CPD_BENCHMARK_FIND", "1") == "1"
     shapes = {
         "resnet18_cifar": ((3, 32, 32), 10),
+        "resnet18_cifar_quant": ((3, 32, 32), 10),
         "resnet50": ((3, 224, 224), 1000),
+        "resnet50_quant": ((3, 224, 224), 1000),
     }
     shape, ncls = shapes[args.model]
     model = build_model(args.model,

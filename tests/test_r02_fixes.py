@@ -133,3 +133,33 @@ def test_draw_curve_svg(tmp_path):
     assert r.returncode == 0, r.stderr
     assert "polyline" in svg.read_text()
     assert len(r.stdout.splitlines()) == 11  # header + 10 rows
+
+
+def test_quantize_model_gemms_conversion():
+    """BASELINE config 4 wiring: conv/linear swapped for Quant_* with weights
+    preserved; (8,23) conversion stays close to the float model (Kahan fp32
+    accumulator vs torch's pairwise sums)."""
+    import torch.nn as nn
+    from cpd_amd.quant import Quant_Conv, Quant_Linear, quantize_model_gemms
+
+    torch.manual_seed(5)
+    m = nn.Sequential(
+        nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(),
+        nn.Conv2d(8, 8, 3, padding=1, groups=8),  # grouped: must NOT convert
+        nn.Flatten(), nn.Linear(8 * 8 * 8, 4))
+    x = torch.randn(2, 3, 8, 8)
+    want = m(x)
+    qm = quantize_model_gemms(m, exp=8, man=23)
+    assert isinstance(qm[0], Quant_Conv)
+    assert isinstance(qm[2], nn.Conv2d)  # grouped conv untouched
+    assert isinstance(qm[4], Quant_Linear)
+    got = qm(x)
+    torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-5)
+
+    # model registry entry builds and steps
+    from cpd_amd.models import build_model
+    rq = build_model("resnet18_cifar_quant", exp=5, man=2)
+    y = rq(torch.randn(2, 3, 32, 32))
+    assert y.shape == (2, 10)
+    y.sum().backward()
+    assert rq.conv1.weight.grad is not None
