@@ -79,12 +79,22 @@ class FusedBNReLU(nn.Module):
         # the counter stays consistent with nn.BatchNorm2d semantics
         if self.training:
             self.num_batches_tracked += 1
+        # channels_last tensors run the native NHWC kernels (C % 4 == 0);
+        # NCHW runs the HW % 4 == 0 kernels.  NEVER .contiguous() a
+        # channels_last tensor here — that would silently transpose to NCHW.
+        nhwc = (x.dim() == 4 and x.shape[1] > 1
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and not x.is_contiguous())
+        shape_ok = (x.shape[1] % 4 == 0) if nhwc else \
+            ((x.shape[2] * x.shape[3]) % 4 == 0)
         use_fused = (self.training and x.is_cuda
-                     and x.dtype == torch.float32
-                     and (x.shape[2] * x.shape[3]) % 4 == 0)
+                     and x.dtype == torch.float32 and shape_ok)
         if not use_fused:
             return self._eager(x, residual)
-        res = residual.contiguous() if residual is not None else None
-        return _FusedBNFn.apply(x.contiguous(), self.weight, self.bias,
-                                self.running_mean, self.running_var,
-                                self.momentum, self.eps, self.relu, res)
+        fmt = torch.channels_last if nhwc else torch.contiguous_format
+        res = residual.contiguous(memory_format=fmt) \
+            if residual is not None else None
+        return _FusedBNFn.apply(x.contiguous(memory_format=fmt), self.weight,
+                                self.bias, self.running_mean,
+                                self.running_var, self.momentum, self.eps,
+                                self.relu, res)

@@ -379,6 +379,226 @@ __global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
   }
 }
 
+// ===========================================================================
+// NHWC (channels_last) variants — x viewed as [R][C], R = N*H*W rows of C
+// contiguous channels.  BN statistics are per COLUMN: lanes own channels
+// (coalesced across the wave), waves stride rows.  This is the layout
+// MIOpen's fast igemm convs want (their NCHW path inserts batched_transpose
+// kernels around every conv — ~9% of the r01 flagship step), so the fused
+// BN must run natively in NHWC for an end-to-end channels_last model.
+// Requires C % 4 == 0 (mask path C % 8 == 0); HW unconstrained.
+// ===========================================================================
+
+__global__ void bn_reduce_nhwc_kernel(const float* __restrict__ x, long R,
+                                      int C, int split,
+                                      float* __restrict__ ws /*[C][split][2]*/) {
+  const int lane = threadIdx.x & 63;
+  const int c = blockIdx.x * 64 + lane;
+  const int wv = threadIdx.x >> 6;  // 4 waves stride rows
+  const int s = blockIdx.y;
+  __shared__ float sm[2][4][64];
+  float sum = 0.f, sq = 0.f;
+  if (c < C) {
+    for (long r = (long)s * 4 + wv; r < R; r += (long)split * 4) {
+      const float v = x[r * C + c];
+      sum += v;
+      sq += v * v;
+    }
+  }
+  sm[0][wv][lane] = sum;
+  sm[1][wv][lane] = sq;
+  __syncthreads();
+  if (wv == 0 && c < C) {
+    for (int w = 1; w < 4; ++w) {
+      sum += sm[0][w][lane];
+      sq += sm[1][w][lane];
+    }
+    ws[((long)c * split + s) * 2 + 0] = sum;
+    ws[((long)c * split + s) * 2 + 1] = sq;
+  }
+}
+
+// fwd normalize (+res)(+relu), 4-wide over channels; per-float4 channel
+// params are contiguous float4 gathers (tiny arrays, L1-resident).
+__global__ void bn_norm_nhwc_kernel(const float* __restrict__ x,
+                                    const float* __restrict__ res,
+                                    float* __restrict__ y, long R, int C,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ gamma,
+                                    const float* __restrict__ beta, int relu) {
+  const long total4 = R * (long)C / 4;
+  const long stride = (long)gridDim.x * TPB;
+  for (long i4 = (long)blockIdx.x * TPB + threadIdx.x; i4 < total4;
+       i4 += stride) {
+    const long i = i4 * 4;
+    const int c0 = (int)(i % C);  // C % 4 == 0 keeps c0..c0+3 in range
+    const float4 mg = ld4(gamma + c0);
+    const float4 mi = ld4(invstd + c0);
+    const float4 mm = ld4(mean + c0);
+    const float4 mb = ld4(beta + c0);
+    const float4 a = {mg.x * mi.x, mg.y * mi.y, mg.z * mi.z, mg.w * mi.w};
+    float4 v = ld4(x + i);
+    v.x = v.x * a.x + (mb.x - mm.x * a.x);
+    v.y = v.y * a.y + (mb.y - mm.y * a.y);
+    v.z = v.z * a.z + (mb.z - mm.z * a.z);
+    v.w = v.w * a.w + (mb.w - mm.w * a.w);
+    if (res) {
+      const float4 r = ld4(res + i);
+      v.x += r.x; v.y += r.y; v.z += r.z; v.w += r.w;
+    }
+    if (relu) {
+      v.x = fmaxf(v.x, 0.f); v.y = fmaxf(v.y, 0.f);
+      v.z = fmaxf(v.z, 0.f); v.w = fmaxf(v.w, 0.f);
+    }
+    st4(y + i, v);
+  }
+}
+
+// fwd normalize with 1-bit/elem ReLU mask (8-wide; C % 8 == 0)
+__global__ void bn_norm_mask_nhwc_kernel(const float* __restrict__ x,
+                                         const float* __restrict__ res,
+                                         float* __restrict__ y,
+                                         unsigned char* __restrict__ mask,
+                                         long R, int C,
+                                         const float* __restrict__ mean,
+                                         const float* __restrict__ invstd,
+                                         const float* __restrict__ gamma,
+                                         const float* __restrict__ beta) {
+  const long total8 = R * (long)C / 8;
+  const long stride = (long)gridDim.x * TPB;
+  for (long i8 = (long)blockIdx.x * TPB + threadIdx.x; i8 < total8;
+       i8 += stride) {
+    const long i = i8 * 8;
+    const int c0 = (int)(i % C);
+    float4 v[2] = {ld4(x + i), ld4(x + i + 4)};
+    unsigned m = 0;
+    for (int h = 0; h < 2; ++h) {
+      const int c = c0 + h * 4;
+      const float4 mg = ld4(gamma + c);
+      const float4 mi = ld4(invstd + c);
+      const float4 mm = ld4(mean + c);
+      const float4 mb = ld4(beta + c);
+      float* pv = reinterpret_cast<float*>(&v[h]);
+      const float* pg = reinterpret_cast<const float*>(&mg);
+      const float* pi = reinterpret_cast<const float*>(&mi);
+      const float* pm = reinterpret_cast<const float*>(&mm);
+      const float* pb = reinterpret_cast<const float*>(&mb);
+      for (int q = 0; q < 4; ++q) {
+        const float aa = pg[q] * pi[q];
+        pv[q] = pv[q] * aa + (pb[q] - pm[q] * aa);
+      }
+    }
+    if (res) {
+      const float4 r0 = ld4(res + i), r1 = ld4(res + i + 4);
+      v[0].x += r0.x; v[0].y += r0.y; v[0].z += r0.z; v[0].w += r0.w;
+      v[1].x += r1.x; v[1].y += r1.y; v[1].z += r1.z; v[1].w += r1.w;
+    }
+    m = (v[0].x > 0.f) | ((v[0].y > 0.f) << 1) | ((v[0].z > 0.f) << 2) |
+        ((v[0].w > 0.f) << 3) | ((v[1].x > 0.f) << 4) |
+        ((v[1].y > 0.f) << 5) | ((v[1].z > 0.f) << 6) | ((v[1].w > 0.f) << 7);
+    v[0].x = fmaxf(v[0].x, 0.f); v[0].y = fmaxf(v[0].y, 0.f);
+    v[0].z = fmaxf(v[0].z, 0.f); v[0].w = fmaxf(v[0].w, 0.f);
+    v[1].x = fmaxf(v[1].x, 0.f); v[1].y = fmaxf(v[1].y, 0.f);
+    v[1].z = fmaxf(v[1].z, 0.f); v[1].w = fmaxf(v[1].w, 0.f);
+    st4(y + i, v[0]);
+    st4(y + i + 4, v[1]);
+    mask[i8] = (unsigned char)m;
+  }
+}
+
+// bwd reduce: per-channel partials of sum(dy_eff), sum(dy_eff * xhat).
+// MASKED selects the 1-bit mask (relu); else dy_eff from y>0 (relu, no
+// mask) or raw dy (y == nullptr).
+__global__ void bn_bwd_reduce_nhwc_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    const float* __restrict__ y, const unsigned char* __restrict__ mask,
+    long R, int C, int split, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ ws) {
+  const int lane = threadIdx.x & 63;
+  const int c = blockIdx.x * 64 + lane;
+  const int wv = threadIdx.x >> 6;
+  const int s = blockIdx.y;
+  __shared__ float sm[2][4][64];
+  float sd = 0.f, sdx = 0.f;
+  if (c < C) {
+    const float m = mean[c];
+    for (long r = (long)s * 4 + wv; r < R; r += (long)split * 4) {
+      const long i = r * C + c;
+      float g = dy[i];
+      if (mask) {
+        g = (mask[i / 8] >> (c & 7)) & 1 ? g : 0.f;  // C%8==0: bit = c&7
+      } else if (y) {
+        g = y[i] > 0.f ? g : 0.f;
+      }
+      sd += g;
+      sdx += g * (x[i] - m);
+    }
+  }
+  sm[0][wv][lane] = sd;
+  sm[1][wv][lane] = sdx;
+  __syncthreads();
+  if (wv == 0 && c < C) {
+    for (int w = 1; w < 4; ++w) {
+      sd += sm[0][w][lane];
+      sdx += sm[1][w][lane];
+    }
+    ws[((long)c * split + s) * 2 + 0] = sd;
+    ws[((long)c * split + s) * 2 + 1] = sdx * invstd[c];
+  }
+}
+
+// bwd dx (+dres), 4-wide over channels (mask path uses 8-wide grouping to
+// read one mask byte per 8 elements)
+__global__ void bn_bwd_dx_nhwc_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    const float* __restrict__ y, const unsigned char* __restrict__ mask,
+    float* __restrict__ dx, float* __restrict__ dres, long R, int C,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ sum_dy,
+    const float* __restrict__ sum_dyx, float inv_count) {
+  const long total4 = R * (long)C / 4;
+  const long stride = (long)gridDim.x * TPB;
+  for (long i4 = (long)blockIdx.x * TPB + threadIdx.x; i4 < total4;
+       i4 += stride) {
+    const long i = i4 * 4;
+    const int c0 = (int)(i % C);
+    float4 g = ld4(dy + i);
+    const float4 v = ld4(x + i);
+    float* pg = reinterpret_cast<float*>(&g);
+    const float* pv = reinterpret_cast<const float*>(&v);
+    if (mask) {
+      const unsigned mk = mask[i / 8];     // C%8==0: byte covers c0&~7..+7
+      const int sh = (int)(i % 8);         // this float4's nibble
+      for (int q = 0; q < 4; ++q)
+        pg[q] = (mk >> (sh + q)) & 1 ? pg[q] : 0.f;
+    } else if (y) {
+      const float4 yy = ld4(y + i);
+      const float* py = reinterpret_cast<const float*>(&yy);
+      for (int q = 0; q < 4; ++q) pg[q] = py[q] > 0.f ? pg[q] : 0.f;
+    }
+    if (dres) st4(dres + i, g);
+    const float4 mm = ld4(mean + c0);
+    const float4 mi = ld4(invstd + c0);
+    const float4 mg = ld4(gamma + c0);
+    const float4 msd = ld4(sum_dy + c0);
+    const float4 msdx = ld4(sum_dyx + c0);
+    const float* pm = reinterpret_cast<const float*>(&mm);
+    const float* pi = reinterpret_cast<const float*>(&mi);
+    const float* pga = reinterpret_cast<const float*>(&mg);
+    const float* psd = reinterpret_cast<const float*>(&msd);
+    const float* psx = reinterpret_cast<const float*>(&msdx);
+    float4 o;
+    float* po = reinterpret_cast<float*>(&o);
+    for (int q = 0; q < 4; ++q) {
+      const float k = pga[q] * pi[q];
+      po[q] = k * (pg[q] - psd[q] * inv_count -
+                   (pv[q] - pm[q]) * pi[q] * psx[q] * inv_count);
+    }
+    st4(dx + i, o);
+  }
+}
+
 // ---------------------------------------------------------------------------
 using at::Tensor;
 
@@ -398,26 +618,47 @@ int pick_split(int N, int C, long HW) {
   return std::min(split, N);
 }
 
+int pick_split_nhwc(long R, int C) {
+  // channel-group blocks are ceil(C/64); split rows so total blocks ~2048
+  const int groups = (C + 63) / 64;
+  long split = 2048 / groups;
+  split = std::min<long>(split, std::max<long>(R / 4, 1));
+  return (int)std::max<long>(split, 1);
+}
+
 std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
                                 const Tensor& beta, Tensor running_mean,
                                 Tensor running_var, double momentum,
                                 double eps, bool relu,
                                 const c10::optional<Tensor>& residual) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous() &&
+  const bool nhwc = x.dim() == 4 && x.size(1) > 1 &&
+                    x.is_contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+              (nhwc || x.is_contiguous()) &&
               x.scalar_type() == at::kFloat);
   const int N = x.size(0), C = x.size(1);
   const long HW = (long)x.size(2) * x.size(3);
-  TORCH_CHECK(HW % 4 == 0, "fused BN needs H*W % 4 == 0");
-  const int split = pick_split(N, C, HW);
+  if (nhwc) {
+    TORCH_CHECK(C % 4 == 0, "fused NHWC BN needs C % 4 == 0");
+  } else {
+    TORCH_CHECK(HW % 4 == 0, "fused BN needs H*W % 4 == 0");
+  }
+  const long R = (long)N * HW;
+  const int split = nhwc ? pick_split_nhwc(R, C) : pick_split(N, C, HW);
   auto opts = x.options();
   Tensor ws = at::empty({C, split, 2}, opts);
   Tensor mean = at::empty({C}, opts);
   Tensor invstd = at::empty({C}, opts);
   Tensor y = at::empty_like(x);
   auto st = cur_stream(x);
-  hipLaunchKernelGGL(bn_reduce_kernel, dim3(C, split), dim3(TPB), 0, st,
-                     x.data_ptr<float>(), N, C, HW, split,
-                     ws.data_ptr<float>());
+  if (nhwc)
+    hipLaunchKernelGGL(bn_reduce_nhwc_kernel, dim3((C + 63) / 64, split),
+                       dim3(TPB), 0, st, x.data_ptr<float>(), R, C, split,
+                       ws.data_ptr<float>());
+  else
+    hipLaunchKernelGGL(bn_reduce_kernel, dim3(C, split), dim3(TPB), 0, st,
+                       x.data_ptr<float>(), N, C, HW, split,
+                       ws.data_ptr<float>());
   hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + TPB - 1) / TPB), dim3(TPB),
                      0, st, ws.data_ptr<float>(), C, split,
                      (float)((long)N * HW), (float)eps, (float)momentum,
@@ -428,19 +669,37 @@ std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
                                            : nullptr);
   const float* res_ptr = nullptr;
   if (residual.has_value()) {
-    TORCH_CHECK(residual->is_contiguous() && residual->sizes() == x.sizes());
+    TORCH_CHECK(residual->sizes() == x.sizes());
+    TORCH_CHECK(nhwc ? residual->is_contiguous(at::MemoryFormat::ChannelsLast)
+                     : residual->is_contiguous(),
+                "residual layout must match x");
     res_ptr = residual->data_ptr<float>();
   }
   Tensor mask = at::empty({0}, opts.dtype(at::kByte));
-  if (relu && HW % 8 == 0) {
+  const bool use_mask = relu && (nhwc ? C % 8 == 0 : HW % 8 == 0);
+  if (use_mask) {
     // emit the ReLU mask as 1 bit/element so backward skips the y re-read
     mask = at::empty({(long)N * C * HW / 8}, opts.dtype(at::kByte));
-    hipLaunchKernelGGL(bn_norm_mask_kernel,
-                       dim3(elem_grid((long)N * C * HW / 8)), dim3(TPB), 0, st,
-                       x.data_ptr<float>(), res_ptr, y.data_ptr<float>(),
-                       mask.data_ptr<uint8_t>(), N, C, HW,
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       gamma.data_ptr<float>(), beta.data_ptr<float>());
+    if (nhwc)
+      hipLaunchKernelGGL(bn_norm_mask_nhwc_kernel,
+                         dim3(elem_grid(R * C / 8)), dim3(TPB), 0, st,
+                         x.data_ptr<float>(), res_ptr, y.data_ptr<float>(),
+                         mask.data_ptr<uint8_t>(), R, C,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), beta.data_ptr<float>());
+    else
+      hipLaunchKernelGGL(bn_norm_mask_kernel,
+                         dim3(elem_grid((long)N * C * HW / 8)), dim3(TPB), 0,
+                         st, x.data_ptr<float>(), res_ptr, y.data_ptr<float>(),
+                         mask.data_ptr<uint8_t>(), N, C, HW,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), beta.data_ptr<float>());
+  } else if (nhwc) {
+    hipLaunchKernelGGL(bn_norm_nhwc_kernel, dim3(elem_grid(R * C / 4)),
+                       dim3(TPB), 0, st, x.data_ptr<float>(), res_ptr,
+                       y.data_ptr<float>(), R, C, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), relu ? 1 : 0);
   } else {
     hipLaunchKernelGGL(bn_norm_kernel, dim3(elem_grid((long)N * C * HW / 4)),
                        dim3(TPB), 0, st, x.data_ptr<float>(), res_ptr,
@@ -456,9 +715,12 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
                                 const Tensor& mean, const Tensor& invstd,
                                 const Tensor& gamma, bool need_dres,
                                 const c10::optional<Tensor>& bitmask) {
+  const bool nhwc = x.dim() == 4 && x.size(1) > 1 &&
+                    x.is_contiguous(at::MemoryFormat::ChannelsLast);
   const int N = x.size(0), C = x.size(1);
   const long HW = (long)x.size(2) * x.size(3);
-  const int split = pick_split(N, C, HW);
+  const long R = (long)N * HW;
+  const int split = nhwc ? pick_split_nhwc(R, C) : pick_split(N, C, HW);
   auto opts = x.options();
   Tensor ws = at::empty({C, split, 2}, opts);
   Tensor sum_dy = at::empty({C}, opts);
@@ -470,11 +732,36 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
   auto st = cur_stream(x);
   const float* yp = y_for_mask.has_value() ? y_for_mask->data_ptr<float>()
                                            : nullptr;
-  Tensor dyc = dy.contiguous();
+  Tensor dyc = nhwc ? dy.contiguous(at::MemoryFormat::ChannelsLast)
+                    : dy.contiguous();
   float* dres_ptr = nullptr;
   if (need_dres) {
     dres = at::empty_like(x);
     dres_ptr = dres.data_ptr<float>();
+  }
+  if (nhwc) {
+    const uint8_t* mk = (bitmask.has_value() && bitmask->defined() &&
+                         bitmask->numel() > 0)
+                            ? bitmask->data_ptr<uint8_t>()
+                            : nullptr;
+    hipLaunchKernelGGL(bn_bwd_reduce_nhwc_kernel, dim3((C + 63) / 64, split),
+                       dim3(TPB), 0, st, x.data_ptr<float>(),
+                       dyc.data_ptr<float>(), mk ? nullptr : yp, mk, R, C,
+                       split, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), ws.data_ptr<float>());
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + TPB - 1) / TPB),
+                       dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
+                       sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
+                       dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
+    hipLaunchKernelGGL(bn_bwd_dx_nhwc_kernel, dim3(elem_grid(R * C / 4)),
+                       dim3(TPB), 0, st, x.data_ptr<float>(),
+                       dyc.data_ptr<float>(), mk ? nullptr : yp, mk,
+                       dx.data_ptr<float>(), dres_ptr, R, C,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                       sum_dyx.data_ptr<float>(), 1.0f / (float)R);
+    if (!need_dres) dres = at::Tensor();
+    return {dx, dgamma, dbeta, dres};
   }
   if (bitmask.has_value() && bitmask->defined() &&
       bitmask->numel() > 0) {

@@ -95,3 +95,79 @@ def test_fused_resnet18_matches_eager_model():
                                   eager.named_parameters()):
         relg = (p1.grad - p2.grad).norm() / p2.grad.norm().clamp_min(1e-8)
         assert relg < 5e-2, (n1, float(relg))
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("relu,with_res,C,HW", [
+    (True, False, 64, (8, 8)), (False, False, 32, (7, 9)),
+    (True, True, 128, (4, 4)), (True, False, 68, (6, 6)),  # C%8!=0: no mask
+])
+def test_fused_bn_nhwc_matches_eager(relu, with_res, C, HW):
+    """channels_last path: native NHWC kernels (no layout transposes) vs
+    the eager NCHW composition."""
+    torch.manual_seed(3)
+    N, (H, W) = 16, HW
+    fmt = torch.channels_last
+    x = torch.randn(N, C, H, W, device="cuda").to(memory_format=fmt)
+    x.requires_grad_(True)
+    res = None
+    if with_res:
+        res = torch.randn(N, C, H, W, device="cuda").to(memory_format=fmt)
+        res.requires_grad_(True)
+
+    m = FusedBNReLU(C, relu=relu).cuda().train()
+    with torch.no_grad():
+        m.weight.mul_(0).add_(torch.rand(C, device="cuda") + 0.5)
+        m.bias.add_(torch.randn(C, device="cuda") * 0.1)
+    ref_bn = torch.nn.BatchNorm2d(C).cuda().train()
+    ref_bn.load_state_dict({k: v for k, v in m.state_dict().items()})
+
+    y = m(x, residual=res)
+    assert y.is_contiguous(memory_format=fmt)
+    xe = x.detach().clone().contiguous().requires_grad_(True)
+    rese = res.detach().clone().contiguous().requires_grad_(True) \
+        if with_res else None
+    ye = ref_bn(xe)
+    if with_res:
+        ye = ye + rese
+    if relu:
+        ye = F.relu(ye)
+    torch.testing.assert_close(y.contiguous(), ye, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(m.running_mean, ref_bn.running_mean,
+                               rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(m.running_var, ref_bn.running_var,
+                               rtol=1e-5, atol=1e-6)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ye.backward(g.contiguous())
+    torch.testing.assert_close(x.grad.contiguous(), xe.grad,
+                               rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(m.weight.grad, ref_bn.weight.grad,
+                               rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(m.bias.grad, ref_bn.bias.grad,
+                               rtol=1e-4, atol=1e-4)
+    if with_res:
+        torch.testing.assert_close(res.grad.contiguous(), rese.grad,
+                                   rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_fused_bn_nhwc_nchw_agree():
+    """Same data through both layout paths: channel sums use the same
+    fixed-shape two-level reduction, so results must agree to fp32
+    round-off; both must be deterministic."""
+    torch.manual_seed(4)
+    N, C, H, W = 32, 64, 16, 16
+    xc = torch.randn(N, C, H, W, device="cuda")
+    outs = {}
+    for fmt in (torch.contiguous_format, torch.channels_last):
+        m = FusedBNReLU(C, relu=True).cuda().train()
+        x = xc.to(memory_format=fmt).requires_grad_(True)
+        y = m(x)
+        y.square().sum().backward()
+        outs[fmt] = (y.detach().contiguous(), x.grad.contiguous(),
+                     m.weight.grad.clone())
+    a, b = outs[torch.contiguous_format], outs[torch.channels_last]
+    torch.testing.assert_close(a[0], b[0], rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(a[1], b[1], rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(a[2], b[2], rtol=1e-4, atol=1e-4)

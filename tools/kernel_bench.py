@@ -89,11 +89,13 @@ def main():
         t = timeit(lambda: a @ b, reps=10)
         report(f"torch.mm (rocBLAS) {sz}^3", t, flops=2 * sz ** 3)
 
-    for sz in (512, 1024):
+    for sz in (512, 1024, 2048):
         a = torch.randn(sz, sz, device=dev)
         b = torch.randn(sz, sz, device=dev)
         t = timeit(lambda: ops.hip_ext().quant_gemm(a, b, 3, 4), reps=5)
         report(f"quant_gemm e4m3 {sz}^3", t, flops=2 * sz ** 3)
+        t = timeit(lambda: ops.hip_ext().quant_gemm(a, b, 2, 5), reps=5)
+        report(f"quant_gemm e5m2 {sz}^3", t, flops=2 * sz ** 3)
 
 
 if __name__ == "__main__":
