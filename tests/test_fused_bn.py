@@ -162,7 +162,8 @@ def test_fused_bn_nhwc_nchw_agree():
     outs = {}
     for fmt in (torch.contiguous_format, torch.channels_last):
         m = FusedBNReLU(C, relu=True).cuda().train()
-        x = xc.to(memory_format=fmt).requires_grad_(True)
+        # clone: .to(memory_format=contiguous_format) would alias xc
+        x = xc.clone(memory_format=fmt).requires_grad_(True)
         y = m(x)
         y.square().sum().backward()
         outs[fmt] = (y.detach().contiguous(), x.grad.contiguous(),

@@ -80,6 +80,34 @@ def main():
     t = timeit(lambda: ops.seg_scale_(flat, offsets, shifts, -1, aligned=True))
     report("seg_scale_ aligned", t, bytes_moved=8 * nb)
 
+    # fused BN fwd+bwd vs eager composition, both layouts, ResNet18/CIFAR
+    # layer shapes (b512)
+    from cpd_amd.models.fused_bn import FusedBNReLU
+
+    def bn_step(m, x):
+        y = m(x)
+        y.backward(gy[x.shape])
+        x.grad = None
+
+    for (N, C, H, W) in [(512, 64, 32, 32), (512, 128, 16, 16),
+                         (512, 256, 8, 8), (512, 512, 4, 4)]:
+        gy = {}
+        for fmt, tag in ((torch.contiguous_format, "nchw"),
+                         (torch.channels_last, "nhwc")):
+            x = torch.randn(N, C, H, W, device=dev).clone(memory_format=fmt)
+            x.requires_grad_(True)
+            gy[x.shape] = torch.randn(N, C, H, W, device=dev).clone(
+                memory_format=fmt)
+            m = FusedBNReLU(C).cuda().train()
+            t = timeit(lambda: bn_step(m, x), reps=10)
+            report(f"fusedBN {tag} {N}x{C}x{H}x{W} fwd+bwd", t,
+                   bytes_moved=(5 + 2) * N * C * H * W * 4)
+            e = torch.nn.Sequential(torch.nn.BatchNorm2d(C),
+                                    torch.nn.ReLU()).cuda().train()
+            t = timeit(lambda: bn_step(e, x), reps=10)
+            report(f"eagerBN {tag} {N}x{C}x{H}x{W} fwd+bwd", t,
+                   bytes_moved=(5 + 2) * N * C * H * W * 4)
+
     # GEMMs
     for sz in (2048, 4096):
         a = torch.randn(sz, sz, device=dev)
