@@ -8,11 +8,15 @@ composition on CPU, in eval mode, or for shapes the kernels don't cover —
 on GPU training the fused path is the one that runs (fails loudly if the HIP
 extension is missing).
 """
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
 from .. import ops
+
+_DEBUG = os.environ.get("CPD_BN_DEBUG") == "1"
 
 
 class _FusedBNFn(torch.autograd.Function):
@@ -89,6 +93,12 @@ class FusedBNReLU(nn.Module):
             ((x.shape[2] * x.shape[3]) % 4 == 0)
         use_fused = (self.training and x.is_cuda
                      and x.dtype == torch.float32 and shape_ok)
+        if _DEBUG and not getattr(self, "_dbg_done", False):
+            self._dbg_done = True
+            print(f"[fused_bn] shape={tuple(x.shape)} nhwc={nhwc} "
+                  f"fused={use_fused} x_contig={x.is_contiguous()} "
+                  f"x_cl={x.is_contiguous(memory_format=torch.channels_last)}",
+                  flush=True)
         if not use_fused:
             return self._eager(x, residual)
         fmt = torch.channels_last if nhwc else torch.contiguous_format
