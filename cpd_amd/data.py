@@ -43,10 +43,19 @@ class ProceduralImages(Dataset):
     so a model that learns the templates generalizes to the val split."""
 
     def __init__(self, n=16384, shape=(3, 32, 32), num_classes=10, seed=0,
-                 templates_per_class=4, snr=0.35):
+                 templates_per_class=4, snr=0.5):
         g = torch.Generator().manual_seed(777)  # templates shared by splits
-        self.templates = torch.randn(
-            (num_classes, templates_per_class) + tuple(shape), generator=g)
+        # LOW-FREQUENCY templates (random 4x4 upsampled bilinearly): a conv
+        # net with pooling can learn these; full-bandwidth white-noise
+        # templates are only learnable by a global matched filter (a linear
+        # probe solves them, conv+avgpool architectures cannot)
+        lowres = torch.randn(
+            (num_classes * templates_per_class, shape[0], 4, 4), generator=g)
+        up = torch.nn.functional.interpolate(
+            lowres, size=shape[1:], mode="bilinear", align_corners=False)
+        up = up / up.std(dim=(1, 2, 3), keepdim=True).clamp_min(1e-6)
+        self.templates = up.view((num_classes, templates_per_class)
+                                 + tuple(shape))
         gs = torch.Generator().manual_seed(seed)
         self.labels = torch.randint(0, num_classes, (n,), generator=gs)
         v = torch.randint(0, templates_per_class, (n,), generator=gs)

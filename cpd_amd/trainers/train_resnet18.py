@@ -60,6 +60,13 @@ def parse_args(argv=None):
                         'experiments without a dataset on disk)')
     p.add_argument('--max_iter', default=0, type=int,
                    help='override the epoch-derived iteration count (>0)')
+    p.add_argument('--peak_lr', default=1.6, type=float,
+                   help='post-warmup peak LR (reference schedule: 1.6 at '
+                        'global batch 4096, mix.py:181-198)')
+    p.add_argument('--warmup_iter', default=0, type=int,
+                   help='override the 5-epoch warmup length in iterations '
+                        '(small procedural datasets make 5 epochs only a '
+                        'few iterations, which diverges)')
     p.add_argument('--data-root', default='./data/cifar-10-batches-py')
     # config-file defaults (res18_cifar.yaml parity)
     p.add_argument('--arch', default='res_cifar')
@@ -80,14 +87,15 @@ def parse_args(argv=None):
     return args
 
 
-def adjust_learning_rate(optimizer, step, iter_per_epoch):
-    """Warmup 0.1 -> 1.6 over 5 epochs, /10 at epochs 40 and 80
-    (mix.py:181-198 schedule)."""
-    warm_up_iter = 5 * iter_per_epoch
+def adjust_learning_rate(optimizer, step, iter_per_epoch, peak_lr=1.6,
+                         warmup_iter=0):
+    """Warmup 0.1 -> peak over 5 epochs (or ``warmup_iter`` iterations),
+    /10 at epochs 40 and 80 (mix.py:181-198 schedule; peak 1.6)."""
+    warm_up_iter = warmup_iter or 5 * iter_per_epoch
     if step <= warm_up_iter:
-        lr = 0.1 + (1.6 - 0.1) * (step / warm_up_iter)
+        lr = 0.1 + (peak_lr - 0.1) * (step / warm_up_iter)
     else:
-        lr = 1.6
+        lr = peak_lr
         if step > iter_per_epoch * 40:
             lr *= 0.1
         if step > iter_per_epoch * 80:
@@ -190,7 +198,9 @@ def train(args, train_loader, val_loader, dm, model, criterion, optimizer,
             curr_step += 1
         if curr_step > max_iter:
             break
-        lr = adjust_learning_rate(optimizer, curr_step, iter_per_epoch)
+        lr = adjust_learning_rate(optimizer, curr_step, iter_per_epoch,
+                                  peak_lr=args.peak_lr,
+                                  warmup_iter=args.warmup_iter)
 
         x = x.to(device, non_blocking=True)
         y = y.to(device, non_blocking=True)

@@ -10,12 +10,16 @@ cd "$GRAFT_REPO_ROOT" || cd /root/repo
 mkdir -p gpurun_out
 export HSA_ENABLE_IPC_MODE_LEGACY=0 MIOPEN_FIND_MODE=FAST
 
-COMMON="--procedural --emulate_node 8 --max_iter 420 --val_freq 30
-        --print_freq 30 --batch_size 512 --workers 2"
+# batch 128 x emulate_node 8 = global 1024 at the reference lr/batch ratio
+# (1.6@4096 -> 0.4@1024); warmup stretched to 80 iters (the 5-epoch rule
+# gives only 16 iters/epoch on the 16k-sample procedural set)
+COMMON="--procedural --emulate_node 8 --max_iter 900 --val_freq 45
+        --print_freq 45 --batch_size 128 --workers 2 --peak_lr 0.4
+        --warmup_iter 80"
 
 run() {  # name extra-flags...
   name=$1; shift
-  timeout 500 python -m cpd_amd.trainers.train_resnet18 $COMMON \
+  timeout 900 python -m cpd_amd.trainers.train_resnet18 $COMMON \
       --save_path gpurun_out/ckpt_$name "$@" \
       > gpurun_out/acc_$name.log 2>&1
   echo "$name rc=$? last:"; grep '\* All Loss' gpurun_out/acc_$name.log | tail -2
