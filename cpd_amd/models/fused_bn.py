@@ -89,8 +89,8 @@ class FusedBNReLU(nn.Module):
         nhwc = (x.dim() == 4 and x.shape[1] > 1
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and not x.is_contiguous())
-        shape_ok = (x.shape[1] % 4 == 0) if nhwc else \
-            ((x.shape[2] * x.shape[3]) % 4 == 0)
+        shape_ok = (x.shape[1] % 4 == 0 and x.shape[1] <= 1024) if nhwc \
+            else ((x.shape[2] * x.shape[3]) % 4 == 0)
         use_fused = (self.training and x.is_cuda
                      and x.dtype == torch.float32 and shape_ok)
         if _DEBUG and not getattr(self, "_dbg_done", False):

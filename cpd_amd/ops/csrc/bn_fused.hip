@@ -381,175 +381,177 @@ __global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
 
 // ===========================================================================
 // NHWC (channels_last) variants — x viewed as [R][C], R = N*H*W rows of C
-// contiguous channels.  BN statistics are per COLUMN: lanes own channels
-// (coalesced across the wave), waves stride rows.  This is the layout
-// MIOpen's fast igemm convs want (their NCHW path inserts batched_transpose
-// kernels around every conv — ~9% of the r01 flagship step), so the fused
-// BN must run natively in NHWC for an end-to-end channels_last model.
-// Requires C % 4 == 0 (mask path C % 8 == 0); HW unconstrained.
+// contiguous channels.  This is the layout MIOpen's fast igemm convs want
+// (their NCHW path inserts batched_transpose kernels around every conv), so
+// the fused BN must run natively in NHWC for an end-to-end channels_last
+// model.
+//
+// v2 geometry (the v1 lane-per-channel scalar loops measured 2.8x slower
+// than the NCHW kernels at C=64): every thread owns one CHANNEL QUAD
+// (q = tid % (C/4)) for its whole lifetime — the per-channel parameters are
+// loaded ONCE into registers, every data access is a float4, consecutive
+// threads cover consecutive quads (perfectly coalesced rows), and there is
+// no 64-bit div/mod in the loop body.  Requires C % 4 == 0 and C <= 1024
+// (mask path C % 8 == 0); HW unconstrained.
 // ===========================================================================
 
+// fwd pass 1: per-channel partial sum/sumsq into ws[C][split][2].
+// grid = (split); one block covers ALL channels (qpc <= 256).
 __global__ void bn_reduce_nhwc_kernel(const float* __restrict__ x, long R,
                                       int C, int split,
-                                      float* __restrict__ ws /*[C][split][2]*/) {
-  const int lane = threadIdx.x & 63;
-  const int c = blockIdx.x * 64 + lane;
-  const int wv = threadIdx.x >> 6;  // 4 waves stride rows
-  const int s = blockIdx.y;
-  __shared__ float sm[2][4][64];
-  float sum = 0.f, sq = 0.f;
-  if (c < C) {
-    for (long r = (long)s * 4 + wv; r < R; r += (long)split * 4) {
-      const float v = x[r * C + c];
-      sum += v;
-      sq += v * v;
+                                      float* __restrict__ ws) {
+  const int qpc = C / 4;
+  const int q = threadIdx.x % qpc;
+  const int r0 = threadIdx.x / qpc;
+  const int rpb = TPB / qpc;
+  __shared__ float4 sm[TPB][2];
+  float4 sum = {0.f, 0.f, 0.f, 0.f}, sq = {0.f, 0.f, 0.f, 0.f};
+  if (r0 < rpb) {
+    for (long r = (long)blockIdx.x * rpb + r0; r < R;
+         r += (long)split * rpb) {
+      const float4 v = ld4(x + r * C + q * 4);
+      sum.x += v.x; sum.y += v.y; sum.z += v.z; sum.w += v.w;
+      sq.x += v.x * v.x; sq.y += v.y * v.y;
+      sq.z += v.z * v.z; sq.w += v.w * v.w;
     }
   }
-  sm[0][wv][lane] = sum;
-  sm[1][wv][lane] = sq;
+  sm[threadIdx.x][0] = sum;
+  sm[threadIdx.x][1] = sq;
   __syncthreads();
-  if (wv == 0 && c < C) {
-    for (int w = 1; w < 4; ++w) {
-      sum += sm[0][w][lane];
-      sq += sm[1][w][lane];
+  if (threadIdx.x < qpc) {
+    for (int g = 1; g < rpb; ++g) {
+      const float4 a = sm[threadIdx.x + g * qpc][0];
+      const float4 b = sm[threadIdx.x + g * qpc][1];
+      sum.x += a.x; sum.y += a.y; sum.z += a.z; sum.w += a.w;
+      sq.x += b.x; sq.y += b.y; sq.z += b.z; sq.w += b.w;
     }
-    ws[((long)c * split + s) * 2 + 0] = sum;
-    ws[((long)c * split + s) * 2 + 1] = sq;
+    const int c0 = threadIdx.x * 4;
+    const float* ps = reinterpret_cast<const float*>(&sum);
+    const float* pq = reinterpret_cast<const float*>(&sq);
+    for (int j = 0; j < 4; ++j) {
+      ws[((long)(c0 + j) * split + blockIdx.x) * 2 + 0] = ps[j];
+      ws[((long)(c0 + j) * split + blockIdx.x) * 2 + 1] = pq[j];
+    }
   }
 }
 
-// fwd normalize (+res)(+relu), 4-wide over channels; per-float4 channel
-// params are contiguous float4 gathers (tiny arrays, L1-resident).
+// fwd pass 2: y = relu(xhat*gamma + beta [+ res]) (+1-bit mask when MASKED)
+template <bool MASKED>
 __global__ void bn_norm_nhwc_kernel(const float* __restrict__ x,
                                     const float* __restrict__ res,
-                                    float* __restrict__ y, long R, int C,
-                                    const float* __restrict__ mean,
+                                    float* __restrict__ y,
+                                    unsigned char* __restrict__ mask, long R,
+                                    int C, const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     const float* __restrict__ gamma,
-                                    const float* __restrict__ beta, int relu) {
-  const long total4 = R * (long)C / 4;
-  const long stride = (long)gridDim.x * TPB;
-  for (long i4 = (long)blockIdx.x * TPB + threadIdx.x; i4 < total4;
-       i4 += stride) {
-    const long i = i4 * 4;
-    const int c0 = (int)(i % C);  // C % 4 == 0 keeps c0..c0+3 in range
-    const float4 mg = ld4(gamma + c0);
-    const float4 mi = ld4(invstd + c0);
-    const float4 mm = ld4(mean + c0);
-    const float4 mb = ld4(beta + c0);
-    const float4 a = {mg.x * mi.x, mg.y * mi.y, mg.z * mi.z, mg.w * mi.w};
-    float4 v = ld4(x + i);
-    v.x = v.x * a.x + (mb.x - mm.x * a.x);
-    v.y = v.y * a.y + (mb.y - mm.y * a.y);
-    v.z = v.z * a.z + (mb.z - mm.z * a.z);
-    v.w = v.w * a.w + (mb.w - mm.w * a.w);
+                                    const float* __restrict__ beta,
+                                    int relu) {
+  const int qpc = C / 4;
+  const int q = threadIdx.x % qpc;
+  const int r0 = threadIdx.x / qpc;
+  const int rpb = TPB / qpc;
+  if (r0 >= rpb) return;
+  const int c0 = q * 4;
+  const float4 mg = ld4(gamma + c0);
+  const float4 mi = ld4(invstd + c0);
+  const float4 mm = ld4(mean + c0);
+  const float4 mb = ld4(beta + c0);
+  const float4 a = {mg.x * mi.x, mg.y * mi.y, mg.z * mi.z, mg.w * mi.w};
+  const float4 bb = {mb.x - mm.x * a.x, mb.y - mm.y * a.y,
+                     mb.z - mm.z * a.z, mb.w - mm.w * a.w};
+  for (long r = (long)blockIdx.x * rpb + r0; r < R;
+       r += (long)gridDim.x * rpb) {
+    const long base = r * C + c0;
+    float4 v = ld4(x + base);
+    v.x = v.x * a.x + bb.x;
+    v.y = v.y * a.y + bb.y;
+    v.z = v.z * a.z + bb.z;
+    v.w = v.w * a.w + bb.w;
     if (res) {
-      const float4 r = ld4(res + i);
-      v.x += r.x; v.y += r.y; v.z += r.z; v.w += r.w;
+      const float4 rr = ld4(res + base);
+      v.x += rr.x; v.y += rr.y; v.z += rr.z; v.w += rr.w;
     }
-    if (relu) {
+    if (MASKED) {
+      // C % 8 == 0: quads pair (even q -> bits 0-3, odd q -> bits 4-7) of
+      // the byte at (r*C + (q & ~1)*4)/8; adjacent tids, same wave
+      unsigned nib = (v.x > 0.f) | ((v.y > 0.f) << 1) | ((v.z > 0.f) << 2) |
+                     ((v.w > 0.f) << 3);
+      const unsigned other = __shfl_down(nib, 1);
+      if ((q & 1) == 0) mask[(r * C) / 8 + q / 2] =
+          (unsigned char)(nib | (other << 4));
+      v.x = fmaxf(v.x, 0.f); v.y = fmaxf(v.y, 0.f);
+      v.z = fmaxf(v.z, 0.f); v.w = fmaxf(v.w, 0.f);
+    } else if (relu) {
       v.x = fmaxf(v.x, 0.f); v.y = fmaxf(v.y, 0.f);
       v.z = fmaxf(v.z, 0.f); v.w = fmaxf(v.w, 0.f);
     }
-    st4(y + i, v);
+    st4(y + base, v);
   }
 }
 
-// fwd normalize with 1-bit/elem ReLU mask (8-wide; C % 8 == 0)
-__global__ void bn_norm_mask_nhwc_kernel(const float* __restrict__ x,
-                                         const float* __restrict__ res,
-                                         float* __restrict__ y,
-                                         unsigned char* __restrict__ mask,
-                                         long R, int C,
-                                         const float* __restrict__ mean,
-                                         const float* __restrict__ invstd,
-                                         const float* __restrict__ gamma,
-                                         const float* __restrict__ beta) {
-  const long total8 = R * (long)C / 8;
-  const long stride = (long)gridDim.x * TPB;
-  for (long i8 = (long)blockIdx.x * TPB + threadIdx.x; i8 < total8;
-       i8 += stride) {
-    const long i = i8 * 8;
-    const int c0 = (int)(i % C);
-    float4 v[2] = {ld4(x + i), ld4(x + i + 4)};
-    unsigned m = 0;
-    for (int h = 0; h < 2; ++h) {
-      const int c = c0 + h * 4;
-      const float4 mg = ld4(gamma + c);
-      const float4 mi = ld4(invstd + c);
-      const float4 mm = ld4(mean + c);
-      const float4 mb = ld4(beta + c);
-      float* pv = reinterpret_cast<float*>(&v[h]);
-      const float* pg = reinterpret_cast<const float*>(&mg);
-      const float* pi = reinterpret_cast<const float*>(&mi);
-      const float* pm = reinterpret_cast<const float*>(&mm);
-      const float* pb = reinterpret_cast<const float*>(&mb);
-      for (int q = 0; q < 4; ++q) {
-        const float aa = pg[q] * pi[q];
-        pv[q] = pv[q] * aa + (pb[q] - pm[q] * aa);
-      }
-    }
-    if (res) {
-      const float4 r0 = ld4(res + i), r1 = ld4(res + i + 4);
-      v[0].x += r0.x; v[0].y += r0.y; v[0].z += r0.z; v[0].w += r0.w;
-      v[1].x += r1.x; v[1].y += r1.y; v[1].z += r1.z; v[1].w += r1.w;
-    }
-    m = (v[0].x > 0.f) | ((v[0].y > 0.f) << 1) | ((v[0].z > 0.f) << 2) |
-        ((v[0].w > 0.f) << 3) | ((v[1].x > 0.f) << 4) |
-        ((v[1].y > 0.f) << 5) | ((v[1].z > 0.f) << 6) | ((v[1].w > 0.f) << 7);
-    v[0].x = fmaxf(v[0].x, 0.f); v[0].y = fmaxf(v[0].y, 0.f);
-    v[0].z = fmaxf(v[0].z, 0.f); v[0].w = fmaxf(v[0].w, 0.f);
-    v[1].x = fmaxf(v[1].x, 0.f); v[1].y = fmaxf(v[1].y, 0.f);
-    v[1].z = fmaxf(v[1].z, 0.f); v[1].w = fmaxf(v[1].w, 0.f);
-    st4(y + i, v[0]);
-    st4(y + i + 4, v[1]);
-    mask[i8] = (unsigned char)m;
-  }
-}
-
-// bwd reduce: per-channel partials of sum(dy_eff), sum(dy_eff * xhat).
-// MASKED selects the 1-bit mask (relu); else dy_eff from y>0 (relu, no
-// mask) or raw dy (y == nullptr).
+// bwd pass 1: per-channel partials of sum(dy_eff), sum(dy_eff * (x-mean)).
 __global__ void bn_bwd_reduce_nhwc_kernel(
     const float* __restrict__ x, const float* __restrict__ dy,
     const float* __restrict__ y, const unsigned char* __restrict__ mask,
     long R, int C, int split, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ ws) {
-  const int lane = threadIdx.x & 63;
-  const int c = blockIdx.x * 64 + lane;
-  const int wv = threadIdx.x >> 6;
-  const int s = blockIdx.y;
-  __shared__ float sm[2][4][64];
-  float sd = 0.f, sdx = 0.f;
-  if (c < C) {
-    const float m = mean[c];
-    for (long r = (long)s * 4 + wv; r < R; r += (long)split * 4) {
-      const long i = r * C + c;
-      float g = dy[i];
+  const int qpc = C / 4;
+  const int q = threadIdx.x % qpc;
+  const int r0 = threadIdx.x / qpc;
+  const int rpb = TPB / qpc;
+  __shared__ float4 sm[TPB][2];
+  float4 sd = {0.f, 0.f, 0.f, 0.f}, sdx = {0.f, 0.f, 0.f, 0.f};
+  const int c0 = q * 4;
+  if (r0 < rpb) {
+    const float4 mm = ld4(mean + c0);
+    for (long r = (long)blockIdx.x * rpb + r0; r < R;
+         r += (long)split * rpb) {
+      const long base = r * C + c0;
+      float4 g = ld4(dy + base);
+      const float4 v = ld4(x + base);
       if (mask) {
-        g = (mask[i / 8] >> (c & 7)) & 1 ? g : 0.f;  // C%8==0: bit = c&7
+        const unsigned mk = mask[(r * C) / 8 + q / 2];
+        const unsigned nib = (q & 1) ? (mk >> 4) : mk;
+        g.x = nib & 1 ? g.x : 0.f;
+        g.y = nib & 2 ? g.y : 0.f;
+        g.z = nib & 4 ? g.z : 0.f;
+        g.w = nib & 8 ? g.w : 0.f;
       } else if (y) {
-        g = y[i] > 0.f ? g : 0.f;
+        const float4 yy = ld4(y + base);
+        g.x = yy.x > 0.f ? g.x : 0.f;
+        g.y = yy.y > 0.f ? g.y : 0.f;
+        g.z = yy.z > 0.f ? g.z : 0.f;
+        g.w = yy.w > 0.f ? g.w : 0.f;
       }
-      sd += g;
-      sdx += g * (x[i] - m);
+      sd.x += g.x; sd.y += g.y; sd.z += g.z; sd.w += g.w;
+      sdx.x += g.x * (v.x - mm.x);
+      sdx.y += g.y * (v.y - mm.y);
+      sdx.z += g.z * (v.z - mm.z);
+      sdx.w += g.w * (v.w - mm.w);
     }
   }
-  sm[0][wv][lane] = sd;
-  sm[1][wv][lane] = sdx;
+  sm[threadIdx.x][0] = sd;
+  sm[threadIdx.x][1] = sdx;
   __syncthreads();
-  if (wv == 0 && c < C) {
-    for (int w = 1; w < 4; ++w) {
-      sd += sm[0][w][lane];
-      sdx += sm[1][w][lane];
+  if (threadIdx.x < qpc) {
+    for (int g = 1; g < rpb; ++g) {
+      const float4 a = sm[threadIdx.x + g * qpc][0];
+      const float4 b = sm[threadIdx.x + g * qpc][1];
+      sd.x += a.x; sd.y += a.y; sd.z += a.z; sd.w += a.w;
+      sdx.x += b.x; sdx.y += b.y; sdx.z += b.z; sdx.w += b.w;
     }
-    ws[((long)c * split + s) * 2 + 0] = sd;
-    ws[((long)c * split + s) * 2 + 1] = sdx * invstd[c];
+    const int cc = threadIdx.x * 4;
+    const float* ps = reinterpret_cast<const float*>(&sd);
+    const float* px = reinterpret_cast<const float*>(&sdx);
+    for (int j = 0; j < 4; ++j) {
+      ws[((long)(cc + j) * split + blockIdx.x) * 2 + 0] = ps[j];
+      ws[((long)(cc + j) * split + blockIdx.x) * 2 + 1] =
+          px[j] * invstd[cc + j];
+    }
   }
 }
 
-// bwd dx (+dres), 4-wide over channels (mask path uses 8-wide grouping to
-// read one mask byte per 8 elements)
+// bwd pass 2: dx (+ dres = dy_eff)
 __global__ void bn_bwd_dx_nhwc_kernel(
     const float* __restrict__ x, const float* __restrict__ dy,
     const float* __restrict__ y, const unsigned char* __restrict__ mask,
@@ -557,45 +559,48 @@ __global__ void bn_bwd_dx_nhwc_kernel(
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ sum_dy,
     const float* __restrict__ sum_dyx, float inv_count) {
-  const long total4 = R * (long)C / 4;
-  const long stride = (long)gridDim.x * TPB;
-  for (long i4 = (long)blockIdx.x * TPB + threadIdx.x; i4 < total4;
-       i4 += stride) {
-    const long i = i4 * 4;
-    const int c0 = (int)(i % C);
-    float4 g = ld4(dy + i);
-    const float4 v = ld4(x + i);
-    float* pg = reinterpret_cast<float*>(&g);
-    const float* pv = reinterpret_cast<const float*>(&v);
+  const int qpc = C / 4;
+  const int q = threadIdx.x % qpc;
+  const int r0 = threadIdx.x / qpc;
+  const int rpb = TPB / qpc;
+  if (r0 >= rpb) return;
+  const int c0 = q * 4;
+  const float4 mm = ld4(mean + c0);
+  const float4 mi = ld4(invstd + c0);
+  const float4 mg = ld4(gamma + c0);
+  const float4 msd = ld4(sum_dy + c0);
+  const float4 msx = ld4(sum_dyx + c0);
+  const float4 k = {mg.x * mi.x, mg.y * mi.y, mg.z * mi.z, mg.w * mi.w};
+  const float4 md = {msd.x * inv_count, msd.y * inv_count,
+                     msd.z * inv_count, msd.w * inv_count};
+  const float4 mx = {msx.x * inv_count * mi.x, msx.y * inv_count * mi.y,
+                     msx.z * inv_count * mi.z, msx.w * inv_count * mi.w};
+  for (long r = (long)blockIdx.x * rpb + r0; r < R;
+       r += (long)gridDim.x * rpb) {
+    const long base = r * C + c0;
+    float4 g = ld4(dy + base);
+    const float4 v = ld4(x + base);
     if (mask) {
-      const unsigned mk = mask[i / 8];     // C%8==0: byte covers c0&~7..+7
-      const int sh = (int)(i % 8);         // this float4's nibble
-      for (int q = 0; q < 4; ++q)
-        pg[q] = (mk >> (sh + q)) & 1 ? pg[q] : 0.f;
+      const unsigned mk = mask[(r * C) / 8 + q / 2];
+      const unsigned nib = (q & 1) ? (mk >> 4) : mk;
+      g.x = nib & 1 ? g.x : 0.f;
+      g.y = nib & 2 ? g.y : 0.f;
+      g.z = nib & 4 ? g.z : 0.f;
+      g.w = nib & 8 ? g.w : 0.f;
     } else if (y) {
-      const float4 yy = ld4(y + i);
-      const float* py = reinterpret_cast<const float*>(&yy);
-      for (int q = 0; q < 4; ++q) pg[q] = py[q] > 0.f ? pg[q] : 0.f;
+      const float4 yy = ld4(y + base);
+      g.x = yy.x > 0.f ? g.x : 0.f;
+      g.y = yy.y > 0.f ? g.y : 0.f;
+      g.z = yy.z > 0.f ? g.z : 0.f;
+      g.w = yy.w > 0.f ? g.w : 0.f;
     }
-    if (dres) st4(dres + i, g);
-    const float4 mm = ld4(mean + c0);
-    const float4 mi = ld4(invstd + c0);
-    const float4 mg = ld4(gamma + c0);
-    const float4 msd = ld4(sum_dy + c0);
-    const float4 msdx = ld4(sum_dyx + c0);
-    const float* pm = reinterpret_cast<const float*>(&mm);
-    const float* pi = reinterpret_cast<const float*>(&mi);
-    const float* pga = reinterpret_cast<const float*>(&mg);
-    const float* psd = reinterpret_cast<const float*>(&msd);
-    const float* psx = reinterpret_cast<const float*>(&msdx);
+    if (dres) st4(dres + base, g);
     float4 o;
-    float* po = reinterpret_cast<float*>(&o);
-    for (int q = 0; q < 4; ++q) {
-      const float k = pga[q] * pi[q];
-      po[q] = k * (pg[q] - psd[q] * inv_count -
-                   (pv[q] - pm[q]) * pi[q] * psx[q] * inv_count);
-    }
-    st4(dx + i, o);
+    o.x = k.x * (g.x - md.x - (v.x - mm.x) * mx.x);
+    o.y = k.y * (g.y - md.y - (v.y - mm.y) * mx.y);
+    o.z = k.z * (g.z - md.z - (v.z - mm.z) * mx.z);
+    o.w = k.w * (g.w - md.w - (v.w - mm.w) * mx.w);
+    st4(dx + base, o);
   }
 }
 
@@ -619,11 +624,16 @@ int pick_split(int N, int C, long HW) {
 }
 
 int pick_split_nhwc(long R, int C) {
-  // channel-group blocks are ceil(C/64); split rows so total blocks ~2048
-  const int groups = (C + 63) / 64;
-  long split = 2048 / groups;
-  split = std::min<long>(split, std::max<long>(R / 4, 1));
-  return (int)std::max<long>(split, 1);
+  // v2: one block covers ALL channels (qpc = C/4 <= 256); split = #blocks
+  const int qpc = C / 4;
+  const int rpb = std::max(TPB / qpc, 1);
+  return (int)std::min<long>(2048, std::max<long>(R / rpb, 1));
+}
+
+inline int nhwc_grid(long R, int C) {
+  const int qpc = C / 4;
+  const int rpb = std::max(TPB / qpc, 1);
+  return (int)std::min<long>((R + rpb - 1) / rpb, 16384);
 }
 
 std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
@@ -639,7 +649,8 @@ std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
   const int N = x.size(0), C = x.size(1);
   const long HW = (long)x.size(2) * x.size(3);
   if (nhwc) {
-    TORCH_CHECK(C % 4 == 0, "fused NHWC BN needs C % 4 == 0");
+    TORCH_CHECK(C % 4 == 0 && C <= 1024,
+                "fused NHWC BN needs C % 4 == 0 and C <= 1024");
   } else {
     TORCH_CHECK(HW % 4 == 0, "fused BN needs H*W % 4 == 0");
   }
@@ -652,8 +663,8 @@ std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
   Tensor y = at::empty_like(x);
   auto st = cur_stream(x);
   if (nhwc)
-    hipLaunchKernelGGL(bn_reduce_nhwc_kernel, dim3((C + 63) / 64, split),
-                       dim3(TPB), 0, st, x.data_ptr<float>(), R, C, split,
+    hipLaunchKernelGGL(bn_reduce_nhwc_kernel, dim3(split), dim3(TPB), 0, st,
+                       x.data_ptr<float>(), R, C, split,
                        ws.data_ptr<float>());
   else
     hipLaunchKernelGGL(bn_reduce_kernel, dim3(C, split), dim3(TPB), 0, st,
@@ -681,12 +692,12 @@ std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
     // emit the ReLU mask as 1 bit/element so backward skips the y re-read
     mask = at::empty({(long)N * C * HW / 8}, opts.dtype(at::kByte));
     if (nhwc)
-      hipLaunchKernelGGL(bn_norm_mask_nhwc_kernel,
-                         dim3(elem_grid(R * C / 8)), dim3(TPB), 0, st,
+      hipLaunchKernelGGL(bn_norm_nhwc_kernel<true>,
+                         dim3(nhwc_grid(R, C)), dim3(TPB), 0, st,
                          x.data_ptr<float>(), res_ptr, y.data_ptr<float>(),
                          mask.data_ptr<uint8_t>(), R, C,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         gamma.data_ptr<float>(), beta.data_ptr<float>());
+                         gamma.data_ptr<float>(), beta.data_ptr<float>(), 1);
     else
       hipLaunchKernelGGL(bn_norm_mask_kernel,
                          dim3(elem_grid((long)N * C * HW / 8)), dim3(TPB), 0,
@@ -695,9 +706,10 @@ std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          gamma.data_ptr<float>(), beta.data_ptr<float>());
   } else if (nhwc) {
-    hipLaunchKernelGGL(bn_norm_nhwc_kernel, dim3(elem_grid(R * C / 4)),
-                       dim3(TPB), 0, st, x.data_ptr<float>(), res_ptr,
-                       y.data_ptr<float>(), R, C, mean.data_ptr<float>(),
+    hipLaunchKernelGGL(bn_norm_nhwc_kernel<false>,
+                       dim3(nhwc_grid(R, C)), dim3(TPB), 0, st,
+                       x.data_ptr<float>(), res_ptr, y.data_ptr<float>(),
+                       (unsigned char*)nullptr, R, C, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                        beta.data_ptr<float>(), relu ? 1 : 0);
   } else {
@@ -744,7 +756,7 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
                          bitmask->numel() > 0)
                             ? bitmask->data_ptr<uint8_t>()
                             : nullptr;
-    hipLaunchKernelGGL(bn_bwd_reduce_nhwc_kernel, dim3((C + 63) / 64, split),
+    hipLaunchKernelGGL(bn_bwd_reduce_nhwc_kernel, dim3(split),
                        dim3(TPB), 0, st, x.data_ptr<float>(),
                        dyc.data_ptr<float>(), mk ? nullptr : yp, mk, R, C,
                        split, mean.data_ptr<float>(),
@@ -753,7 +765,7 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
                        dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
                        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
-    hipLaunchKernelGGL(bn_bwd_dx_nhwc_kernel, dim3(elem_grid(R * C / 4)),
+    hipLaunchKernelGGL(bn_bwd_dx_nhwc_kernel, dim3(nhwc_grid(R, C)),
                        dim3(TPB), 0, st, x.data_ptr<float>(),
                        dyc.data_ptr<float>(), mk ? nullptr : yp, mk,
                        dx.data_ptr<float>(), dres_ptr, R, C,
