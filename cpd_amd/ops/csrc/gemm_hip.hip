@@ -212,7 +212,9 @@ constexpr int QBM = 32, QBN = 32, QBK = 16;
 __global__ __launch_bounds__(256) void quant_gemm_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     float* __restrict__ C, int M, int N, int K, int man, int exp) {
-  __shared__ float As[QBK][QBM + 1];
+  // row pad +2 keeps the per-lane (2m, 2m+1) operand pairs 8-byte aligned
+  // so each a/b pair is one ds_read_b64
+  __shared__ float As[QBK][QBM + 2];
   __shared__ float Bs[QBK][QBN];
 
   const int tx = threadIdx.x & 15;   // 16x16 threads, 2x2 outputs each
@@ -247,11 +249,10 @@ __global__ __launch_bounds__(256) void quant_gemm_kernel(
 
     const int klim = min(QBK, K - k0);  // never round in padded-k steps
     for (int kk = 0; kk < klim; ++kk) {  // strictly k-ordered (semantics)
-      float a[2], b[2];
-      a[0] = As[kk][ty * 2];
-      a[1] = As[kk][ty * 2 + 1];
-      b[0] = Bs[kk][tx * 2];
-      b[1] = Bs[kk][tx * 2 + 1];
+      const float2 a01 = *reinterpret_cast<const float2*>(&As[kk][ty * 2]);
+      const float2 b01 = *reinterpret_cast<const float2*>(&Bs[kk][tx * 2]);
+      const float a[2] = {a01.x, a01.y};
+      const float b[2] = {b01.x, b01.y};
       for (int i = 0; i < 2; ++i)
         for (int j = 0; j < 2; ++j) {
           // cast_fp_fast: ~20-VALU float-pipeline cast, bit-identical to
