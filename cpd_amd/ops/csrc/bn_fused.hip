@@ -67,6 +67,9 @@ __global__ void bn_reduce_kernel(const float* __restrict__ x, int N, int C,
 }
 
 // ---- forward pass 1b: finalize mean/invstd + update running stats ---------
+// One WAVE per channel (lanes stride the split axis): with the NHWC reduce
+// using up to 2048 splits, the r01 one-THREAD-per-channel loop serialized
+// C*split*2 workspace reads and dominated the whole BN (kernel_bench r02).
 __global__ void bn_finalize_kernel(const float* __restrict__ ws, int C,
                                    int split, float count, float eps,
                                    float momentum,
@@ -74,13 +77,19 @@ __global__ void bn_finalize_kernel(const float* __restrict__ ws, int C,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var) {
-  const int c = blockIdx.x * TPB + threadIdx.x;
+  const int c = blockIdx.x * (TPB / 64) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
   if (c >= C) return;
   float sum = 0.f, sumsq = 0.f;
-  for (int s = 0; s < split; ++s) {
+  for (int s = lane; s < split; s += 64) {
     sum += ws[((long)c * split + s) * 2 + 0];
     sumsq += ws[((long)c * split + s) * 2 + 1];
   }
+  for (int d = 32; d > 0; d >>= 1) {
+    sum += __shfl_down(sum, d);
+    sumsq += __shfl_down(sumsq, d);
+  }
+  if (lane != 0) return;
   const float m = sum / count;
   const float var = fmaxf(sumsq / count - m * m, 0.0f);
   mean[c] = m;
@@ -324,13 +333,19 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ ws, int C,
                                        float* __restrict__ sum_dyx,
                                        float* __restrict__ dgamma,
                                        float* __restrict__ dbeta) {
-  const int c = blockIdx.x * TPB + threadIdx.x;
+  const int c = blockIdx.x * (TPB / 64) + (threadIdx.x >> 6);  // wave/channel
+  const int lane = threadIdx.x & 63;
   if (c >= C) return;
   float sd = 0.f, sdx = 0.f;
-  for (int s = 0; s < split; ++s) {
+  for (int s = lane; s < split; s += 64) {
     sd += ws[((long)c * split + s) * 2 + 0];
     sdx += ws[((long)c * split + s) * 2 + 1];
   }
+  for (int d = 32; d > 0; d >>= 1) {
+    sd += __shfl_down(sd, d);
+    sdx += __shfl_down(sdx, d);
+  }
+  if (lane != 0) return;
   sum_dy[c] = sd;
   sum_dyx[c] = sdx;
   dbeta[c] = sd;
@@ -624,10 +639,14 @@ int pick_split(int N, int C, long HW) {
 }
 
 int pick_split_nhwc(long R, int C) {
-  // v2: one block covers ALL channels (qpc = C/4 <= 256); split = #blocks
+  // one block covers ALL channels (qpc = C/4 <= 256); split = #blocks.
+  // Cap C*split so the finalize pass stays small (wave-per-channel reads
+  // split entries per lane-stride).
   const int qpc = C / 4;
   const int rpb = std::max(TPB / qpc, 1);
-  return (int)std::min<long>(2048, std::max<long>(R / rpb, 1));
+  long split = std::min<long>(2048, std::max<long>(R / rpb, 1));
+  split = std::min<long>(split, std::max<long>(524288 / C, 64));
+  return (int)split;
 }
 
 inline int nhwc_grid(long R, int C) {
@@ -670,8 +689,8 @@ std::vector<Tensor> bn_relu_fwd(const Tensor& x, const Tensor& gamma,
     hipLaunchKernelGGL(bn_reduce_kernel, dim3(C, split), dim3(TPB), 0, st,
                        x.data_ptr<float>(), N, C, HW, split,
                        ws.data_ptr<float>());
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + TPB - 1) / TPB), dim3(TPB),
-                     0, st, ws.data_ptr<float>(), C, split,
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + TPB / 64 - 1) / (TPB / 64)),
+                     dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
                      (float)((long)N * HW), (float)eps, (float)momentum,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      running_mean.defined() ? running_mean.data_ptr<float>()
@@ -761,8 +780,9 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
                        dyc.data_ptr<float>(), mk ? nullptr : yp, mk, R, C,
                        split, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), ws.data_ptr<float>());
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + TPB - 1) / TPB),
-                       dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel,
+                       dim3((C + TPB / 64 - 1) / (TPB / 64)), dim3(TPB), 0,
+                       st, ws.data_ptr<float>(), C, split,
                        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
     hipLaunchKernelGGL(bn_bwd_dx_nhwc_kernel, dim3(nhwc_grid(R, C)),
@@ -782,8 +802,9 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
                        0, st, x.data_ptr<float>(), dyc.data_ptr<float>(), mk,
                        N, C, HW, split, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), ws.data_ptr<float>());
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + TPB - 1) / TPB),
-                       dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel,
+                       dim3((C + TPB / 64 - 1) / (TPB / 64)), dim3(TPB), 0,
+                       st, ws.data_ptr<float>(), C, split,
                        sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
     hipLaunchKernelGGL(bn_bwd_dx_mask_kernel,
@@ -801,8 +822,9 @@ std::vector<Tensor> bn_relu_bwd(const Tensor& x, const Tensor& dy,
                      x.data_ptr<float>(), dyc.data_ptr<float>(), yp, N, C, HW,
                      split, mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      ws.data_ptr<float>());
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + TPB - 1) / TPB),
-                     dim3(TPB), 0, st, ws.data_ptr<float>(), C, split,
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel,
+                     dim3((C + TPB / 64 - 1) / (TPB / 64)), dim3(TPB), 0, st,
+                     ws.data_ptr<float>(), C, split,
                      sum_dy.data_ptr<float>(), sum_dyx.data_ptr<float>(),
                      dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(elem_grid((long)N * C * HW / 4)),
