@@ -194,12 +194,22 @@ __global__ void seg_maxabs_kernel(const float* __restrict__ x, long n,
   const int lane = threadIdx.x & 63;
   const long nwin = (n + 255) >> 8;
   const long wstride = (long)gridDim.x * (TPB / 64);
+  // monotonic per-wave segment hint: windows advance left-to-right, so the
+  // next window's segment is found by a short forward scan instead of a
+  // fresh binary search of dependent global loads per window
+  int s0 = -1;
+  long s0_end = -1;
   for (long w = (long)blockIdx.x * (TPB / 64) + (threadIdx.x >> 6); w < nwin;
        w += wstride) {
     const long lo = w << 8;
     const long hi = min(lo + 256, n);
-    const int s0 = find_seg(ofs, S, lo);
-    if (hi == lo + 256 && ofs[s0 + 1] >= hi) {
+    if (s0 < 0) {
+      s0 = find_seg(ofs, S, lo);
+      s0_end = ofs[s0 + 1];
+    } else {
+      while (s0_end <= lo) s0_end = ofs[++s0 + 1];
+    }
+    if (hi == lo + 256 && s0_end >= hi) {
       const float4 v = reinterpret_cast<const float4*>(x + lo)[lane];
       float lmax = fmaxf(fmaxf(fabsf(v.x), fabsf(v.y)),
                          fmaxf(fabsf(v.z), fabsf(v.w)));
@@ -305,12 +315,19 @@ __global__ void scale_quantize_kernel(float* __restrict__ x, long n,
   const int lane = threadIdx.x & 63;
   const long nwin = (n + 255) >> 8;
   const long wstride = (long)gridDim.x * (TPB / 64);
+  int s0 = -1;
+  long s0_end = -1;  // monotonic per-wave hint (see seg_maxabs_kernel)
   for (long w = (long)blockIdx.x * (TPB / 64) + (threadIdx.x >> 6); w < nwin;
        w += wstride) {
     const long lo = w << 8;
     const long hi = min(lo + 256, n);
-    const int s0 = find_seg(ofs, S, lo);
-    if (hi == lo + 256 && ofs[s0 + 1] >= hi) {
+    if (s0 < 0) {
+      s0 = find_seg(ofs, S, lo);
+      s0_end = ofs[s0 + 1];
+    } else {
+      while (s0_end <= lo) s0_end = ofs[++s0 + 1];
+    }
+    if (hi == lo + 256 && s0_end >= hi) {
       const float scale =
           ldexpf(1.0f, (int)shifts[s0] * (sign_only ? sign_only : 1));
       float4* p = reinterpret_cast<float4*>(x + lo);
