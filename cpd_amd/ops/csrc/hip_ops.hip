@@ -196,9 +196,15 @@ __global__ void seg_maxabs_kernel(const float* __restrict__ x, long n,
   const long wstride = (long)gridDim.x * (TPB / 64);
   // monotonic per-wave segment hint: windows advance left-to-right, so the
   // next window's segment is found by a short forward scan instead of a
-  // fresh binary search of dependent global loads per window
+  // fresh binary search per window.  The per-lane max ACCUMULATES across
+  // consecutive same-segment windows and flushes (wave-reduce + one atomic)
+  // only on segment change — one atomic per window serialized badly on the
+  // S counters (0.15 TB/s measured r02); the launcher caps the grid so each
+  // wave owns many windows.
   int s0 = -1;
   long s0_end = -1;
+  int acc_seg = -1;
+  float lmax = 0.0f;
   for (long w = (long)blockIdx.x * (TPB / 64) + (threadIdx.x >> 6); w < nwin;
        w += wstride) {
     const long lo = w << 8;
@@ -210,19 +216,31 @@ __global__ void seg_maxabs_kernel(const float* __restrict__ x, long n,
       while (s0_end <= lo) s0_end = ofs[++s0 + 1];
     }
     if (hi == lo + 256 && s0_end >= hi) {
+      if (s0 != acc_seg) {
+        if (acc_seg >= 0) {  // flush the previous segment's accumulator
+          for (int d = 32; d > 0; d >>= 1)
+            lmax = fmaxf(lmax, __shfl_xor(lmax, d));
+          if (lane == 0 && f32_bits(lmax) != 0)
+            atomicMax(out_bits + acc_seg, f32_bits(lmax));
+        }
+        acc_seg = s0;
+        lmax = 0.0f;
+      }
       const float4 v = reinterpret_cast<const float4*>(x + lo)[lane];
-      float lmax = fmaxf(fmaxf(fabsf(v.x), fabsf(v.y)),
-                         fmaxf(fabsf(v.z), fabsf(v.w)));
-      for (int d = 32; d > 0; d >>= 1)
-        lmax = fmaxf(lmax, __shfl_xor(lmax, d));
-      if (lane == 0 && f32_bits(lmax) != 0)
-        atomicMax(out_bits + s0, f32_bits(lmax));
+      lmax = fmaxf(lmax, fmaxf(fmaxf(fabsf(v.x), fabsf(v.y)),
+                               fmaxf(fabsf(v.z), fabsf(v.w))));
     } else {
       for (long j = lo + lane; j < hi; j += 64) {
         const int s = find_seg(ofs, S, j);
         atomicMax(out_bits + s, f32_bits(fabsf(x[j])));
       }
     }
+  }
+  if (acc_seg >= 0) {
+    for (int d = 32; d > 0; d >>= 1)
+      lmax = fmaxf(lmax, __shfl_xor(lmax, d));
+    if (lane == 0 && f32_bits(lmax) != 0)
+      atomicMax(out_bits + acc_seg, f32_bits(lmax));
   }
 }
 
@@ -472,8 +490,11 @@ Tensor seg_max_exp(const Tensor& flat, const Tensor& offsets,
                        kSegPerBlock / 4);
   } else if (n) {
     const long nwin = (n + 255) >> 8;
+    // cap at 1024 blocks (4096 waves saturate HBM) so each wave accumulates
+    // across many windows -> ~1 atomic per (wave, segment) instead of per
+    // window
     const int wblocks =
-        (int)std::min<long>((nwin + TPB / 64 - 1) / (TPB / 64), 16384);
+        (int)std::min<long>((nwin + TPB / 64 - 1) / (TPB / 64), 1024);
     hipLaunchKernelGGL(seg_maxabs_kernel, dim3(wblocks), dim3(TPB), 0,
                        cur_stream(flat), flat.data_ptr<float>(), n,
                        offsets.data_ptr<int64_t>(), S,
