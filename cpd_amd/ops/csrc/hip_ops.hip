@@ -205,8 +205,13 @@ __global__ void seg_maxabs_kernel(const float* __restrict__ x, long n,
   long s0_end = -1;
   int acc_seg = -1;
   float lmax = 0.0f;
-  for (long w = (long)blockIdx.x * (TPB / 64) + (threadIdx.x >> 6); w < nwin;
-       w += wstride) {
+  // contiguous per-wave window range: keeps loads streaming and the
+  // segment hint advancing through each segment once (strided windows put
+  // consecutive windows ~MBs apart — a serial hint-walk per window)
+  const long wid = (long)blockIdx.x * (TPB / 64) + (threadIdx.x >> 6);
+  const long per = (nwin + wstride - 1) / wstride;
+  const long wlo = wid * per, wcap = min(wlo + per, nwin);
+  for (long w = wlo; w < wcap; ++w) {
     const long lo = w << 8;
     const long hi = min(lo + 256, n);
     if (s0 < 0) {
@@ -335,8 +340,10 @@ __global__ void scale_quantize_kernel(float* __restrict__ x, long n,
   const long wstride = (long)gridDim.x * (TPB / 64);
   int s0 = -1;
   long s0_end = -1;  // monotonic per-wave hint (see seg_maxabs_kernel)
-  for (long w = (long)blockIdx.x * (TPB / 64) + (threadIdx.x >> 6); w < nwin;
-       w += wstride) {
+  const long wid = (long)blockIdx.x * (TPB / 64) + (threadIdx.x >> 6);
+  const long per = (nwin + wstride - 1) / wstride;
+  const long wlo = wid * per, wcap = min(wlo + per, nwin);
+  for (long w = wlo; w < wcap; ++w) {
     const long lo = w << 8;
     const long hi = min(lo + 256, n);
     if (s0 < 0) {
