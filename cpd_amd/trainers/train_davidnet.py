@@ -145,7 +145,7 @@ def run_epoch(args, loader, model, step, optimizer, lr_sched, epoch, device,
                     g['lr'] = lr
                 step.substep(loss * args.loss_scale /
                              step.loss_scale_denom())
-            tot_loss += float(loss)
+            tot_loss += float(loss.detach())
             tot_correct += float(out['correct'].sum())
             n += x.shape[0]
             steps += 1

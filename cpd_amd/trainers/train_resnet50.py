@@ -155,7 +155,7 @@ def train_epoch(args, loader, model, criterion, optimizer, step, device,
             out = model(x)
             loss = criterion(out, y) / step.loss_scale_denom()
             step.substep(loss)
-        losses.update(float(loss) * step.loss_scale_denom())
+        losses.update(float(loss.detach()) * step.loss_scale_denom())
         top1.update(float(accuracy(out, y)[0]))
         if rank == 0 and (s + 1) % 50 == 0:
             img_s = (s + 1) * args.batch_size * args.batches_per_allreduce * \
