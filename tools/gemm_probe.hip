@@ -295,6 +295,91 @@ __global__ __launch_bounds__(256) void gemm_v5(const float* __restrict__ A,
     }
 }
 
+// ---- v6: pair-interleave A ONLY (B stays row-major float4-write): the v5
+// PMC run showed 67M LDS bank conflicts from B's scattered interleaved
+// writes; A's transposed write was already scalar so interleaving it is
+// free, and the A reads become ds_read_b64.
+__global__ __launch_bounds__(256) void gemm_v6(const float* __restrict__ A,
+                                               const float* __restrict__ B,
+                                               float* __restrict__ C, int M,
+                                               int N, int K) {
+  __shared__ float As[2][BK][BM + 2];
+  __shared__ float Bs[2][BK][BN];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64, wc = (wave & 1) * 64;
+  const int tid = threadIdx.x;
+  const int nwg = gridDim.x * gridDim.y;
+  const int wg = blockIdx.y * gridDim.x + blockIdx.x;
+  const int q = nwg / 8, rr = nwg % 8, xcd = wg % 8, idx = wg / 8;
+  const int swg = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  const int block_row = (swg % gridDim.x) * BM;
+  const int block_col = (swg / gridDim.x) * BN;
+
+  f32x16 acc[2][2] = {};
+  const int ktiles = K / BK;
+  StageRegs regs;
+  auto write6 = [&](float (*as)[BM + 2], float (*bs)[BN]) {
+    const int k4 = tid & 7, m0 = tid >> 3;
+    for (int p = 0; p < 4; ++p) {
+      const int pm = m0 * 2 + (p & 1) + (p >> 1) * 64;
+      as[k4 * 4 + 0][pm] = regs.a[p].x;
+      as[k4 * 4 + 1][pm] = regs.a[p].y;
+      as[k4 * 4 + 2][pm] = regs.a[p].z;
+      as[k4 * 4 + 3][pm] = regs.a[p].w;
+    }
+    const int n4 = tid & 31, kk0 = tid >> 5;
+    for (int p = 0; p < 4; ++p)
+      *reinterpret_cast<float4*>(&bs[kk0 + p * 8][n4 * 4]) = regs.b[p];
+  };
+  stage_load(A, B, K, N, block_row, block_col, 0, tid, regs);
+  write6(As[0], Bs[0]);
+  int cur = 0;
+  const int l31 = lane & 31, kh = lane >> 5;
+  const int ga = (wr >> 6) * 64 + l31 * 2;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < ktiles)
+      stage_load(A, B, K, N, block_row, block_col, (kt + 1) * BK, tid, regs);
+    for (int kk = 0; kk < BK; kk += 2) {
+      const float2 a01 = *reinterpret_cast<const float2*>(&As[cur][kk + kh][ga]);
+      const float b0 = Bs[cur][kk + kh][wc + l31];
+      const float b1 = Bs[cur][kk + kh][wc + 32 + l31];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.x, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.x, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.y, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.y, b1, acc[1][1], 0, 0, 0);
+    }
+    if (kt + 1 < ktiles) write6(As[cur ^ 1], Bs[cur ^ 1]);
+    cur ^= 1;
+  }
+  for (int mi = 0; mi < 2; ++mi)
+    for (int nj = 0; nj < 2; ++nj) {
+      const int col = block_col + wc + nj * 32 + (lane & 31);
+      for (int r = 0; r < 16; ++r) {
+        const int row = block_row + wr + mi * 32 + (r & 3) + 8 * (r >> 2) +
+                        4 * (lane >> 5);
+        C[(long)row * N + col] = acc[mi][nj][r];
+      }
+    }
+}
+
+double bench_v6(const float* dA, const float* dB, float* dC, int Nsz,
+                int reps) {
+  dim3 grid(Nsz / BM, Nsz / BN), block(256);
+  hipLaunchKernelGGL(gemm_v6, grid, block, 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+  HIP_CHECK(hipDeviceSynchronize());
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0); hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < reps; ++i)
+    hipLaunchKernelGGL(gemm_v6, grid, block, 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+  hipEventRecord(t1);
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms; hipEventElapsedTime(&ms, t0, t1);
+  return 2.0 * Nsz * Nsz * (double)Nsz * reps / (ms * 1e-3) / 1e12;
+}
+
 double bench_v5(const float* dA, const float* dB, float* dC, int Nsz,
                 int reps) {
   dim3 grid(Nsz / BM, Nsz / BN), block(256);
@@ -476,11 +561,19 @@ int main(int argc, char** argv) {
     for (long i = 0; i < (long)Nsz * Nsz; ++i) bad += (h0[i] != h1[i]);
     printf("v5check: %ld mismatches\n", bad);
     if (bad) return 1;
+    hipLaunchKernelGGL(gemm_v6, g0, dim3(256), 0, 0, dA, dB, dC, Nsz, Nsz, Nsz);
+    HIP_CHECK(hipMemcpy(h1.data(), dC, (long)Nsz * Nsz * 4, hipMemcpyDeviceToHost));
+    bad = 0;
+    for (long i = 0; i < (long)Nsz * Nsz; ++i) bad += (h0[i] != h1[i]);
+    printf("v6check: %ld mismatches\n", bad);
+    if (bad) return 1;
   }
   for (int round = 0; round < rounds; ++round) {
-    printf("round %d: v0=%6.1f v1=%6.1f v5pair=%6.1f big=%6.1f TF\n", round,
+    printf("round %d: v0=%6.1f v1=%6.1f v5pair=%6.1f v6apair=%6.1f big=%6.1f TF\n",
+           round,
            bench<0>(dA, dB, dC, Nsz, reps), bench<1>(dA, dB, dC, Nsz, reps),
-           bench_v5(dA, dB, dC, Nsz, reps), bench_big(dA, dB, dC, Nsz, reps));
+           bench_v5(dA, dB, dC, Nsz, reps), bench_v6(dA, dB, dC, Nsz, reps),
+           bench_big(dA, dB, dC, Nsz, reps));
   }
   return 0;
 }
