@@ -67,6 +67,8 @@ def parse_args(argv=None):
                    help='override the 5-epoch warmup length in iterations '
                         '(small procedural datasets make 5 epochs only a '
                         'few iterations, which diverges)')
+    p.add_argument('--snr', default=0.5, type=float,
+                   help='procedural-dataset template-to-noise ratio')
     p.add_argument('--data-root', default='./data/cifar-10-batches-py')
     # config-file defaults (res18_cifar.yaml parity)
     p.add_argument('--arch', default='res_cifar')
@@ -133,8 +135,8 @@ def main(argv=None):
 
     if args.procedural:
         from cpd_amd.data import ProceduralImages
-        train_set = ProceduralImages(16384, seed=0)
-        val_set = ProceduralImages(2048, seed=1)
+        train_set = ProceduralImages(16384, seed=0, snr=args.snr)
+        val_set = ProceduralImages(2048, seed=1, snr=args.snr)
     elif args.synthetic or not os.path.isdir(args.data_root):
         if not args.synthetic and rank == 0:
             print(f'No CIFAR at {args.data_root}; using synthetic data.')
