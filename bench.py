@@ -92,8 +92,14 @@ def main():
     use_gpu = torch.cuda.is_available() and args.device != "cpu"
     device = torch.device(args.device or ("cuda" if use_gpu else "cpu"))
 
-    if world > 1:
+    # init the process group whenever launched via torchrun (WORLD_SIZE in
+    # the env), including W=1: a single-rank torchrun run then validates
+    # RCCL init / barrier / all_reduce on hardware (plain `python bench.py`
+    # stays group-free)
+    if world > 1 or "WORLD_SIZE" in os.environ:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29517")
+        os.environ.setdefault("RANK", "0")
         local = int(os.environ.get("LOCAL_RANK", rank))
         if use_gpu:
             torch.cuda.set_device(local)
@@ -194,7 +200,7 @@ def main():
     def sync():
         if use_gpu:
             torch.cuda.synchronize()
-        if world > 1:
+        if dist.is_available() and dist.is_initialized():
             dist.barrier()
         if use_gpu:
             torch.cuda.synchronize()
@@ -263,7 +269,7 @@ def main():
         }
         print(json.dumps(result), flush=True)
 
-    if world > 1:
+    if dist.is_available() and dist.is_initialized():
         dist.destroy_process_group()
 
 
